@@ -1,0 +1,207 @@
+"""TensorServingClient — the user-facing API (L5).
+
+Method-for-method parity with reference requests.py:22-110 (same method
+names, kwargs, defaults — timeout 60 s for inference, 10 s for status;
+``model_spec.version.value`` via Int64Value), plus:
+
+* ``get_model_metadata_request`` / ``multi_inference_request`` (the two
+  PredictionService rpcs the reference never wrapped),
+* torch tensors (CPU or CUDA) accepted anywhere a numpy array is,
+* ``use_tensor_content`` encode switch (default True = memcpy fast path),
+* one stub per channel instead of a new stub per request (the reference
+  constructs a stub per call — requests.py:40 — a deliberate fix here),
+* device tensors route through the HIP pack path in ``ops`` when available.
+"""
+from __future__ import annotations
+
+from typing import Any, Dict, Optional, Union
+
+import grpc
+import numpy as np
+
+try:
+    import torch
+except ImportError:  # pragma: no cover
+    torch = None
+
+from .tensors import tensor_to_tensor_proto, tensor_proto_to_ndarray
+from .wire import messages as pb
+from .wire.grpc_stubs import ModelServiceStub, PredictionServiceStub
+
+TensorLike = Union[np.ndarray, "torch.Tensor"]
+
+
+class TensorServingClient:
+    """gRPC client for a TF-Serving-compatible PredictionService."""
+
+    def __init__(self, host: str, port: int,
+                 credentials: Optional[grpc.ChannelCredentials] = None,
+                 options: Optional[list] = None) -> None:
+        self._host_address = f"{host}:{port}"
+        default_options = [
+            ("grpc.max_send_message_length", 1 << 30),
+            ("grpc.max_receive_message_length", 1 << 30),
+        ]
+        opts = default_options + (options or [])
+        if credentials:
+            self._channel = grpc.secure_channel(
+                self._host_address, credentials, options=opts)
+        else:
+            self._channel = grpc.insecure_channel(
+                self._host_address, options=opts)
+        self._prediction_stub = PredictionServiceStub(self._channel)
+        self._model_stub = ModelServiceStub(self._channel)
+
+    # ------------------------------------------------------------------
+    def close(self) -> None:
+        self._channel.close()
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *exc):
+        self.close()
+
+    # ------------------------------------------------------------------
+    def _fill_model_spec(self, spec, model_name: str,
+                         model_version: Optional[int],
+                         signature_name: Optional[str] = None) -> None:
+        spec.name = model_name
+        if model_version is not None:
+            spec.version.value = model_version
+        if signature_name:
+            spec.signature_name = signature_name
+
+    def _encode_input(self, value: TensorLike, use_tensor_content: bool):
+        if torch is not None and isinstance(value, torch.Tensor) \
+                and value.is_cuda:
+            from . import ops
+            return ops.pack_tensor_proto(value)
+        return tensor_to_tensor_proto(value, use_tensor_content)
+
+    # -- Predict --------------------------------------------------------
+    def predict_request(
+        self,
+        model_name: str,
+        input_dict: Dict[str, TensorLike],
+        timeout: int = 60,
+        model_version: Optional[int] = None,
+        signature_name: Optional[str] = None,
+        output_filter: Optional[list] = None,
+        use_tensor_content: bool = True,
+    ) -> "pb.PredictResponse":
+        request = pb.PredictRequest()
+        self._fill_model_spec(request.model_spec, model_name, model_version,
+                              signature_name)
+        for k, v in input_dict.items():
+            request.inputs[k].CopyFrom(
+                self._encode_input(v, use_tensor_content))
+        if output_filter:
+            request.output_filter.extend(output_filter)
+        return self._prediction_stub.Predict(request, timeout)
+
+    # -- Classify / Regress ---------------------------------------------
+    # The reference routes classification/regression through stub.Predict
+    # with a Predict{Request,Response} (requests.py:40,49) — its
+    # classification_request/regression_request only differ in annotation.
+    # We keep that call-compatible behavior (dict-of-tensors in, Predict
+    # rpc) and additionally expose true Classify/Regress rpcs below.
+    def classification_request(
+        self,
+        model_name: str,
+        input_dict: Dict[str, TensorLike],
+        timeout: int = 60,
+        model_version: Optional[int] = None,
+    ):
+        return self.predict_request(model_name, input_dict, timeout,
+                                    model_version)
+
+    def regression_request(
+        self,
+        model_name: str,
+        input_dict: Dict[str, TensorLike],
+        timeout: int = 60,
+        model_version: Optional[int] = None,
+    ):
+        return self.predict_request(model_name, input_dict, timeout,
+                                    model_version)
+
+    def classify(self, model_name: str, input_proto: "pb.Input",
+                 timeout: int = 60,
+                 model_version: Optional[int] = None
+                 ) -> "pb.ClassificationResponse":
+        """True Classify rpc over tensorflow.serving.Input examples."""
+        request = pb.ClassificationRequest()
+        self._fill_model_spec(request.model_spec, model_name, model_version)
+        request.input.CopyFrom(input_proto)
+        return self._prediction_stub.Classify(request, timeout)
+
+    def regress(self, model_name: str, input_proto: "pb.Input",
+                timeout: int = 60,
+                model_version: Optional[int] = None
+                ) -> "pb.RegressionResponse":
+        request = pb.RegressionRequest()
+        self._fill_model_spec(request.model_spec, model_name, model_version)
+        request.input.CopyFrom(input_proto)
+        return self._prediction_stub.Regress(request, timeout)
+
+    # -- Model status / metadata / admin --------------------------------
+    def model_status_request(
+        self,
+        model_name: str,
+        model_version: Optional[int] = None,
+        timeout: Optional[int] = 10,
+    ) -> "pb.GetModelStatusResponse":
+        request = pb.GetModelStatusRequest()
+        request.model_spec.name = model_name
+        if model_version:
+            request.model_spec.version.value = model_version
+        return self._model_stub.GetModelStatus(request, timeout)
+
+    def get_model_metadata_request(
+        self,
+        model_name: str,
+        model_version: Optional[int] = None,
+        metadata_fields: Optional[list] = None,
+        timeout: Optional[int] = 10,
+    ) -> "pb.GetModelMetadataResponse":
+        request = pb.GetModelMetadataRequest()
+        self._fill_model_spec(request.model_spec, model_name, model_version)
+        request.metadata_field.extend(metadata_fields or ["signature_def"])
+        return self._prediction_stub.GetModelMetadata(request, timeout)
+
+    def multi_inference_request(
+        self,
+        tasks: list,
+        input_proto: "pb.Input",
+        timeout: int = 60,
+    ) -> "pb.MultiInferenceResponse":
+        """tasks: list of (model_name, method_name[, model_version])."""
+        request = pb.MultiInferenceRequest()
+        for task in tasks:
+            t = request.tasks.add()
+            name, method = task[0], task[1]
+            version = task[2] if len(task) > 2 else None
+            self._fill_model_spec(t.model_spec, name, version)
+            t.method_name = method
+        request.input.CopyFrom(input_proto)
+        return self._prediction_stub.MultiInference(request, timeout)
+
+    def reload_config_request(self, model_configs: Dict[str, str],
+                              timeout: Optional[int] = 10
+                              ) -> "pb.ReloadConfigResponse":
+        """model_configs: {model_name: base_path}."""
+        request = pb.ReloadConfigRequest()
+        cfg_list = request.config.model_config_list
+        for name, base_path in model_configs.items():
+            c = cfg_list.config.add()
+            c.name = name
+            c.base_path = base_path
+            c.model_platform = "tensorflow"
+        return self._model_stub.HandleReloadConfigRequest(request, timeout)
+
+
+def decode_predict_response(response,
+                            as_numpy: bool = True) -> Dict[str, Any]:
+    """Convenience: PredictResponse -> {name: ndarray}."""
+    return {k: tensor_proto_to_ndarray(v) for k, v in response.outputs.items()}

