@@ -1,0 +1,259 @@
+#!/usr/bin/env python3
+"""Flagship serving benchmark: Predict req/s + p50 round-trip.
+
+Headline config (BASELINE.json): 32x3x224x224 fp32 PredictRequest per GPU
+per step against a local loopback PredictionService, full round trip —
+HIP/staging pack of the device tensor into wire bytes, gRPC over a unix
+socket to a separate server process, C++ parse + echo, response unpacked
+back to HBM. For --gpus N > 1 (torchrun, one rank per GPU) the step is the
+data-parallel config 4: rank 0 scatters the global batch over RCCL/xGMI,
+each rank round-trips its shard, responses are all-gathered.
+
+Contract: rank 0 prints ONE JSON line; value is the WHOLE-JOB req/s
+aggregate; timing brackets exactly K steps between barrier+synchronize
+pairs; MAX elapsed over ranks.
+
+Usage:
+  python bench.py                         # 1 GPU (or CPU fallback), quick
+  python bench.py --gpus 8 --steps 64     # under torch.distributed.run
+  python bench.py --bench-config bert     # BASELINE config 3
+  python bench.py --encoding proto        # python-protobuf client (A/B)
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import multiprocessing
+import os
+import statistics
+import sys
+import time
+
+_ROOT = os.path.dirname(os.path.abspath(__file__))
+if _ROOT not in sys.path:
+    sys.path.insert(0, _ROOT)
+
+import numpy as np  # noqa: E402
+import torch  # noqa: E402
+
+
+def _server_proc(address: str, ready, stop):
+    """Loopback PredictionService in its own process (own GIL)."""
+    if _ROOT not in sys.path:
+        sys.path.insert(0, _ROOT)
+    from min_tfs_client_amd.server import ModelServer, identity_servable
+    with ModelServer(address=address, raw_predict=True, max_workers=8) as srv:
+        srv.manager.load("default", identity_servable(), version=1)
+        ready.set()
+        stop.wait()
+
+
+def make_inputs(cfg: str, device, batch_override=None):
+    if cfg == "resnet50":
+        b = batch_override or 32
+        return {"images": torch.randn(b, 3, 224, 224, device=device,
+                                      dtype=torch.float32)}
+    if cfg == "bert":
+        b = batch_override or 128
+        g = torch.Generator(device="cpu").manual_seed(0)
+        ids = torch.randint(0, 30522, (b, 512), generator=g,
+                            dtype=torch.int32).to(device)
+        mask = torch.ones(b, 512, dtype=torch.int32, device=device)
+        return {"input_ids": ids, "attention_mask": mask}
+    if cfg == "scalar":
+        return {"x": torch.zeros((), dtype=torch.float32, device=device)}
+    if cfg == "bf16pack":
+        b = batch_override or 32
+        return {"images": torch.randn(b, 3, 224, 224, device=device,
+                                      dtype=torch.bfloat16)}
+    raise ValueError(f"unknown bench config {cfg}")
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int,
+                    default=int(os.environ.get("WORLD_SIZE", "1")))
+    ap.add_argument("--steps", type=int, default=64)
+    ap.add_argument("--warmup", type=int, default=16)
+    ap.add_argument("--bench-config", default="resnet50",
+                    choices=["resnet50", "bert", "scalar", "bf16pack"])
+    ap.add_argument("--encoding", default="turbo",
+                    choices=["turbo", "proto"],
+                    help="turbo = C++ codec raw-bytes path; proto = "
+                         "python-protobuf client (reference-style)")
+    ap.add_argument("--copy-mode", type=int, default=0,
+                    help="0 = pinned-staged pipelined copies, 1 = direct "
+                         "pageable hipMemcpy (A/B)")
+    ap.add_argument("--transport", default="unix", choices=["unix", "tcp"])
+    args = ap.parse_args()
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    n_gpus = max(args.gpus, world_size)
+
+    has_gpu = torch.cuda.is_available()
+    if has_gpu:
+        torch.cuda.set_device(local_rank)
+        device = torch.device(f"cuda:{local_rank}")
+    else:
+        device = torch.device("cpu")
+
+    dist = None
+    if world_size > 1:
+        import torch.distributed as dist_mod
+        dist = dist_mod
+        backend = "nccl" if has_gpu else "gloo"
+        dist.init_process_group(backend=backend)
+
+    # ---- per-rank loopback server ------------------------------------
+    if args.transport == "unix":
+        address = f"unix:///tmp/mi355x_bench_{os.getpid()}_{rank}.sock"
+    else:
+        address = None
+    ctx = multiprocessing.get_context("spawn")
+    ready, stop = ctx.Event(), ctx.Event()
+    if address is None:
+        # tcp: pick port in parent to pass to child
+        import socket as _socket
+        s = _socket.socket()
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+        s.close()
+        address = f"127.0.0.1:{port}"
+    proc = ctx.Process(target=_server_proc, args=(address, ready, stop),
+                       daemon=True)
+    proc.start()
+    if not ready.wait(60):
+        raise RuntimeError("bench server failed to start")
+
+    # ---- client ------------------------------------------------------
+    inputs = make_inputs(args.bench_config, device)
+    per_rank_batch = next(iter(inputs.values())).shape[0] \
+        if next(iter(inputs.values())).dim() > 0 else 1
+
+    if args.encoding == "turbo":
+        from min_tfs_client_amd.turbo import TurboPredictClient
+        client = TurboPredictClient(address)
+
+        def step_fn(step_inputs):
+            out_dev = f"cuda:{local_rank}" if has_gpu else "cpu"
+            return client.predict("default", step_inputs,
+                                  output_device=out_dev,
+                                  copy_mode=args.copy_mode)
+    else:
+        from min_tfs_client_amd.client import TensorServingClient
+        host, port = address.split("//")[-1], None
+        if args.transport == "unix":
+            raise SystemExit("--encoding proto requires --transport tcp")
+        host, port = address.split(":")
+        client = TensorServingClient(host, int(port))
+
+        def step_fn(step_inputs):
+            resp = client.predict_request("default", step_inputs)
+            from min_tfs_client_amd.tensors import tensor_proto_to_ndarray
+            return {k: tensor_proto_to_ndarray(v)
+                    for k, v in resp.outputs.items()}
+
+    # ---- one data-parallel step --------------------------------------
+    # world_size > 1 (config 4): rank0's global batch is scattered over
+    # RCCL, each rank round-trips its shard, responses all-gathered.
+    keys = sorted(inputs.keys())
+
+    def dp_step():
+        step_inputs = {}
+        if dist is not None:
+            for k in keys:
+                full = inputs[k]
+                shard = torch.empty_like(full)
+                if rank == 0:
+                    scatter_list = [full.contiguous()
+                                    for _ in range(world_size)]
+                else:
+                    scatter_list = None
+                dist.scatter(shard, scatter_list, src=0)
+                step_inputs[k] = shard
+        else:
+            step_inputs = inputs
+        outs = step_fn(step_inputs)
+        if dist is not None:
+            for k, v in outs.items():
+                if isinstance(v, torch.Tensor) and v.dim() > 0:
+                    gathered = [torch.empty_like(v)
+                                for _ in range(world_size)]
+                    dist.all_gather(gathered, v.contiguous())
+        return outs
+
+    def sync():
+        if has_gpu:
+            torch.cuda.synchronize()
+        if dist is not None:
+            dist.barrier()
+
+    # ---- warmup ------------------------------------------------------
+    for _ in range(args.warmup):
+        dp_step()
+    sync()
+
+    # ---- timed region ------------------------------------------------
+    lat = []
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        s0 = time.perf_counter()
+        dp_step()
+        lat.append(time.perf_counter() - s0)
+    sync()
+    t1 = time.perf_counter()
+
+    elapsed = t1 - t0
+    if dist is not None:
+        e = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if has_gpu else "cpu")
+        dist.all_reduce(e, op=dist.ReduceOp.MAX)
+        elapsed = float(e.item())
+
+    # whole-job aggregate: one request per rank per step
+    reqs_per_s = n_gpus * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1e3
+    p50_ms = statistics.median(lat) * 1e3
+
+    if rank == 0:
+        result = {
+            "metric": "predict_req_per_s",
+            "value": round(reqs_per_s, 3),
+            "unit": "req/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "p50_ms_rtt": round(p50_ms, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": ("bf16" if args.bench_config == "bf16pack"
+                      else "int32" if args.bench_config == "bert"
+                      else "fp32"),
+            "data": "synthetic",
+            "config": {
+                "model": f"identity-echo[{args.bench_config}]",
+                "global_batch": per_rank_batch * n_gpus,
+                "seq_len": 512 if args.bench_config == "bert" else None,
+                "shape_per_request": list(
+                    next(iter(inputs.values())).shape),
+                "parallelism": f"dp{n_gpus}",
+                "encoding": args.encoding,
+                "copy_mode": args.copy_mode,
+                "transport": args.transport,
+                "gpu": has_gpu,
+            },
+        }
+        print(json.dumps(result))
+
+    stop.set()
+    proc.join(timeout=10)
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -91,7 +91,9 @@ def identity_servable() -> Servable:
             out[name] = v
         return out
 
-    return Servable(fn, signature={"method_name": "tensorflow/serving/predict"})
+    s = Servable(fn, signature={"method_name": "tensorflow/serving/predict"})
+    s.is_identity = True  # enables the all-C++ echo fast path in raw mode
+    return s
 
 
 class _Version:
@@ -390,14 +392,72 @@ class ModelServiceImpl(ModelServiceServicer):
 # Server wrapper
 # ---------------------------------------------------------------------------
 
+def _raw_predict_handler(manager: ModelManager, device: str,
+                         metrics: MetricsRegistry):
+    """Raw-bytes Predict: request bytes in, response bytes out, all codec
+    work in the C++ extension (turbo path — see turbo.py)."""
+    from .ops import require_native
+
+    def raw_predict(data, context):
+        t0 = time.perf_counter()
+        native = require_native()
+        try:
+            spec, inputs, _filter = native.parse_predict_request(
+                data, device, 0)
+        except Exception as e:  # noqa: BLE001
+            _abort(context, grpc.StatusCode.INVALID_ARGUMENT,
+                   f"request parsing error: {e}")
+        version = spec["version"] if spec["version"] >= 0 else None
+        try:
+            servable = manager.get(spec["name"], version)
+        except KeyError as e:
+            _abort(context, grpc.StatusCode.NOT_FOUND, str(e))
+        if getattr(servable, "is_identity", False):
+            out = native.echo_predict(data)
+            metrics.observe_request("predict", time.perf_counter() - t0)
+            return out
+        try:
+            outputs = servable(inputs)
+        except Exception as e:  # noqa: BLE001
+            _abort(context, grpc.StatusCode.INTERNAL, str(e))
+        names = list(outputs.keys())
+        tensors = []
+        for k in names:
+            v = outputs[k]
+            if not isinstance(v, torch.Tensor):
+                v = torch.as_tensor(np.asarray(v))
+            tensors.append(v)
+        blob = native.serialize_predict_response(
+            spec["name"], -1 if version is None else version,
+            spec["signature_name"] or "serving_default", names, tensors, 0)
+        metrics.observe_request("predict", time.perf_counter() - t0)
+        return blob
+
+    def identity(x):
+        return x
+
+    return grpc.unary_unary_rpc_method_handler(
+        raw_predict, request_deserializer=identity,
+        response_serializer=identity)
+
+
 class ModelServer:
     """Build-and-start wrapper (Server::BuildAndStart analogue,
-    reference server.cc:291-339)."""
+    reference server.cc:291-339).
+
+    ``raw_predict=True`` swaps the Predict method onto the C++ codec
+    (identity (de)serializers; turbo clients and standard protobuf clients
+    both interoperate — the wire bytes are identical). ``address`` may be a
+    "unix:///path.sock" target for low-overhead loopback serving.
+    """
 
     def __init__(self, port: int = 0, max_workers: int = 16,
                  output_encoding: str = "tensor_content",
                  manager: Optional[ModelManager] = None,
-                 servable_factory=None):
+                 servable_factory=None,
+                 raw_predict: bool = False,
+                 device: str = "cpu",
+                 address: Optional[str] = None):
         self.manager = manager or ModelManager()
         self.metrics = MetricsRegistry()
         self._server = grpc.server(
@@ -407,10 +467,51 @@ class ModelServer:
         self.prediction_service = PredictionServiceImpl(
             self.manager, output_encoding, self.metrics)
         self.model_service = ModelServiceImpl(self.manager, servable_factory)
-        add_PredictionServiceServicer_to_server(self.prediction_service,
-                                                self._server)
-        add_ModelServiceServicer_to_server(self.model_service, self._server)
-        self.port = self._server.add_insecure_port(f"127.0.0.1:{port}")
+        if raw_predict:
+            from .wire import messages as _pb
+            from .wire.grpc_stubs import (
+                add_ModelServiceServicer_to_server as _add_ms)
+            handlers = {
+                "Predict": _raw_predict_handler(self.manager, device,
+                                                self.metrics),
+                "Classify": grpc.unary_unary_rpc_method_handler(
+                    self.prediction_service.Classify,
+                    request_deserializer=_pb.ClassificationRequest.FromString,
+                    response_serializer=(
+                        _pb.ClassificationResponse.SerializeToString)),
+                "Regress": grpc.unary_unary_rpc_method_handler(
+                    self.prediction_service.Regress,
+                    request_deserializer=_pb.RegressionRequest.FromString,
+                    response_serializer=(
+                        _pb.RegressionResponse.SerializeToString)),
+                "MultiInference": grpc.unary_unary_rpc_method_handler(
+                    self.prediction_service.MultiInference,
+                    request_deserializer=(
+                        _pb.MultiInferenceRequest.FromString),
+                    response_serializer=(
+                        _pb.MultiInferenceResponse.SerializeToString)),
+                "GetModelMetadata": grpc.unary_unary_rpc_method_handler(
+                    self.prediction_service.GetModelMetadata,
+                    request_deserializer=(
+                        _pb.GetModelMetadataRequest.FromString),
+                    response_serializer=(
+                        _pb.GetModelMetadataResponse.SerializeToString)),
+            }
+            self._server.add_generic_rpc_handlers((
+                grpc.method_handlers_generic_handler(
+                    "tensorflow.serving.PredictionService", handlers),))
+            _add_ms(self.model_service, self._server)
+        else:
+            add_PredictionServiceServicer_to_server(self.prediction_service,
+                                                    self._server)
+            add_ModelServiceServicer_to_server(self.model_service,
+                                               self._server)
+        if address is not None:
+            self.address = address
+            self.port = self._server.add_insecure_port(address)
+        else:
+            self.port = self._server.add_insecure_port(f"127.0.0.1:{port}")
+            self.address = f"127.0.0.1:{self.port}"
 
     def start(self) -> "ModelServer":
         self._server.start()

@@ -1,0 +1,38 @@
+"""In-tree build of the native extension (gfx950).
+
+    python setup.py build_ext --inplace
+
+Produces ``min_tfs_client_amd/_native*.so`` next to the package so the
+built artefact travels with the repo snapshot to GPU boxes (build system
+analogue of reference setup.py:28-98, which compiles protos instead — our
+wire layer needs no codegen, the native build is the codec + HIP kernels).
+"""
+import os
+
+from setuptools import setup
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+
+_CSRC = os.path.join("min_tfs_client_amd", "ops", "csrc")
+
+ext = CUDAExtension(
+    name="min_tfs_client_amd._native",
+    sources=[
+        os.path.join(_CSRC, "native.cpp"),
+        os.path.join(_CSRC, "pack_kernels.hip"),
+    ],
+    extra_compile_args={
+        "cxx": ["-O3", "-std=c++17"],
+        "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
+    },
+)
+
+setup(
+    name="min-tfs-client-amd",
+    version="0.1.0",
+    packages=["min_tfs_client_amd"],
+    ext_modules=[ext],
+    cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
+)
