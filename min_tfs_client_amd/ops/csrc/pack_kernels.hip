@@ -152,31 +152,47 @@ __global__ void cast_kernel(const typename CVT::In* __restrict__ in,
 // ---------------------------------------------------------------------------
 // fused NCHW->NHWC + cast
 // ---------------------------------------------------------------------------
-// Small-C specialization (C <= 8, the image case C=3): each lane owns 4
-// consecutive hw positions and ALL channels. Reads are coalesced per channel
-// stream; writes are fully coalesced (each lane writes 4*C consecutive
-// outputs). One pass, no LDS needed.
-template <typename CVT, int MAXC>
+// Small-C specialization (the image case C=3): each lane owns HWPL
+// consecutive hw positions and ALL channels. Per-channel reads are 16 B
+// vector loads (G13) when the span is full and aligned; each lane writes
+// HWPL*C consecutive outputs -> fully coalesced wide stores.
+template <typename CVT, int MAXC, int HWPL>
 __global__ void nchw_nhwc_smallc_kernel(
     const typename CVT::In* __restrict__ in,
     typename CVT::Out* __restrict__ out,
     int C, int64_t HW, int64_t N) {
+  using In = typename CVT::In;
   using Out = typename CVT::Out;
-  const int64_t hw_quads = (HW + 3) / 4;
+  const int64_t hw_chunks = (HW + HWPL - 1) / HWPL;
   int64_t idx0 = int64_t(blockIdx.x) * blockDim.x + threadIdx.x;
   int64_t stride = int64_t(gridDim.x) * blockDim.x;
-  for (int64_t idx = idx0; idx < N * hw_quads; idx += stride) {
-    const int64_t nimg = idx / hw_quads;
-    const int64_t hw0 = (idx - nimg * hw_quads) * 4;
+  for (int64_t idx = idx0; idx < N * hw_chunks; idx += stride) {
+    const int64_t nimg = idx / hw_chunks;
+    const int64_t hw0 = (idx - nimg * hw_chunks) * HWPL;
     const int64_t in_base = nimg * C * HW;
     const int64_t out_base = nimg * HW * C;
-    Out vout[4 * MAXC];
-    const int span = int(min(int64_t(4), HW - hw0));
+    Out vout[HWPL * MAXC];
+    const int span = int(min(int64_t(HWPL), HW - hw0));
+    const bool full = span == HWPL;
 #pragma unroll
     for (int c = 0; c < MAXC; ++c) {
       if (c >= C) break;
-      for (int k = 0; k < span; ++k) {
-        vout[k * C + c] = CVT::cvt(in[in_base + c * HW + hw0 + k]);
+      const In* src = in + in_base + c * HW + hw0;
+      In vin[HWPL];
+      if (full &&
+          (reinterpret_cast<uintptr_t>(src) & 15) == 0 &&
+          sizeof(In) * HWPL % 16 == 0) {
+        constexpr int NV = int(sizeof(In) * HWPL / 16);
+#pragma unroll
+        for (int v = 0; v < NV; ++v)
+          reinterpret_cast<int4*>(vin)[v] =
+              reinterpret_cast<const int4*>(src)[v];
+      } else {
+        for (int k = 0; k < span; ++k) vin[k] = src[k];
+      }
+#pragma unroll
+      for (int k = 0; k < HWPL; ++k) {
+        if (k < span) vout[k * C + c] = CVT::cvt(vin[k]);
       }
     }
     Out* dst = out + out_base + hw0 * C;
@@ -315,11 +331,22 @@ static void launch_cast(const at::Tensor& in, at::Tensor& out) {
 template <typename CVT>
 static void launch_nchw_nhwc(const at::Tensor& in, at::Tensor& out,
                              int64_t N, int64_t C, int64_t HW) {
-  if (C <= 8) {
+  if (C <= 4) {
     const int block = 256;
-    const int64_t hw_quads = (HW + 3) / 4;
-    const int grid = grid_for(N * hw_quads, block);
-    hipLaunchKernelGGL((nchw_nhwc_smallc_kernel<CVT, 8>), dim3(grid),
+    const int64_t hw_chunks = (HW + 7) / 8;
+    const int grid = grid_for(N * hw_chunks, block);
+    hipLaunchKernelGGL((nchw_nhwc_smallc_kernel<CVT, 4, 8>), dim3(grid),
+                       dim3(block), 0, current_stream(),
+                       reinterpret_cast<const typename CVT::In*>(
+                           in.const_data_ptr()),
+                       reinterpret_cast<typename CVT::Out*>(
+                           out.mutable_data_ptr()),
+                       int(C), HW, N);
+  } else if (C <= 8) {
+    const int block = 256;
+    const int64_t hw_chunks = (HW + 3) / 4;
+    const int grid = grid_for(N * hw_chunks, block);
+    hipLaunchKernelGGL((nchw_nhwc_smallc_kernel<CVT, 8, 4>), dim3(grid),
                        dim3(block), 0, current_stream(),
                        reinterpret_cast<const typename CVT::In*>(
                            in.const_data_ptr()),

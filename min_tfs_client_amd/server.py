@@ -458,6 +458,8 @@ class ModelServer:
                  raw_predict: bool = False,
                  device: str = "cpu",
                  address: Optional[str] = None):
+        from .utils.allocator import tune_malloc
+        tune_malloc()
         self.manager = manager or ModelManager()
         self.metrics = MetricsRegistry()
         self._server = grpc.server(

@@ -26,6 +26,7 @@ except ImportError:  # pragma: no cover
     torch = None
 
 from .ops import require_native
+from .utils.allocator import tune_malloc
 from .utils.metrics import MetricsRegistry
 
 _PREDICT_PATH = "/tensorflow.serving.PredictionService/Predict"
@@ -51,6 +52,7 @@ class TurboPredictClient:
                  credentials: Optional[grpc.ChannelCredentials] = None,
                  options: Optional[list] = None):
         self._native = require_native()
+        tune_malloc()  # large wire buffers: arena reuse, no per-call mmap
         opts = _CHANNEL_OPTS + (options or [])
         if credentials:
             self._channel = grpc.secure_channel(target, credentials,
