@@ -85,6 +85,10 @@ def main():
                     help="0 = pinned-staged pipelined copies, 1 = direct "
                          "pageable hipMemcpy (A/B)")
     ap.add_argument("--transport", default="unix", choices=["unix", "tcp"])
+    ap.add_argument("--pipeline", type=int, default=1,
+                    help="in-flight requests per rank (1 = sequential; "
+                         ">1 overlaps serialize/transport/parse of "
+                         "consecutive requests)")
     args = ap.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -190,18 +194,52 @@ def main():
         if dist is not None:
             dist.barrier()
 
+    if args.pipeline > 1 and world_size > 1:
+        raise SystemExit("--pipeline > 1 is single-rank only")
+    if args.pipeline > 1 and args.encoding != "turbo":
+        raise SystemExit("--pipeline > 1 requires --encoding turbo")
+
+    def run_pipelined(nsteps):
+        """Sliding window of `pipeline` in-flight requests; returns
+        per-request submit->complete latencies."""
+        out_dev = f"cuda:{local_rank}" if has_gpu else "cpu"
+        lat_local = []
+        inflight = []
+        submitted = 0
+        while submitted < min(args.pipeline, nsteps):
+            fut, dec = client.predict_future("default", inputs)
+            inflight.append((fut, dec, time.perf_counter()))
+            submitted += 1
+        done = 0
+        while done < nsteps:
+            fut, dec, ts = inflight.pop(0)
+            dec(fut.result(), output_device=out_dev)
+            lat_local.append(time.perf_counter() - ts)
+            done += 1
+            if submitted < nsteps:
+                fut, dec = client.predict_future("default", inputs)
+                inflight.append((fut, dec, time.perf_counter()))
+                submitted += 1
+        return lat_local
+
     # ---- warmup ------------------------------------------------------
-    for _ in range(args.warmup):
-        dp_step()
+    if args.pipeline > 1:
+        run_pipelined(args.warmup)
+    else:
+        for _ in range(args.warmup):
+            dp_step()
     sync()
 
     # ---- timed region ------------------------------------------------
     lat = []
     t0 = time.perf_counter()
-    for _ in range(args.steps):
-        s0 = time.perf_counter()
-        dp_step()
-        lat.append(time.perf_counter() - s0)
+    if args.pipeline > 1:
+        lat = run_pipelined(args.steps)
+    else:
+        for _ in range(args.steps):
+            s0 = time.perf_counter()
+            dp_step()
+            lat.append(time.perf_counter() - s0)
     sync()
     t1 = time.perf_counter()
 
@@ -241,6 +279,7 @@ def main():
                 "shape_per_request": list(
                     next(iter(inputs.values())).shape),
                 "parallelism": f"dp{n_gpus}",
+                "pipeline": args.pipeline,
                 "encoding": args.encoding,
                 "copy_mode": args.copy_mode,
                 "transport": args.transport,

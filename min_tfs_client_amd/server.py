@@ -465,7 +465,8 @@ class ModelServer:
         self._server = grpc.server(
             futures.ThreadPoolExecutor(max_workers=max_workers),
             options=[("grpc.max_send_message_length", 1 << 30),
-                     ("grpc.max_receive_message_length", 1 << 30)])
+                     ("grpc.max_receive_message_length", 1 << 30),
+                     ("grpc.http2.max_frame_size", 16 * 1024 * 1024 - 1)])
         self.prediction_service = PredictionServiceImpl(
             self.manager, output_encoding, self.metrics)
         self.model_service = ModelServiceImpl(self.manager, servable_factory)

@@ -34,6 +34,9 @@ _PREDICT_PATH = "/tensorflow.serving.PredictionService/Predict"
 _CHANNEL_OPTS = [
     ("grpc.max_send_message_length", 1 << 30),
     ("grpc.max_receive_message_length", 1 << 30),
+    # 16MB HTTP/2 frames: measured -12% RTT / -21% pipelined ms-per-req on
+    # the 19MB Predict payload vs the 16KB default (tools/bench_grpc_ab.py)
+    ("grpc.http2.max_frame_size", 16 * 1024 * 1024 - 1),
 ]
 
 
