@@ -1,0 +1,86 @@
+"""Data-parallel sharding tests on CPU: 2-process gloo group, scatter ->
+local predict -> all-gather (the config-4 path without GPUs; on MI355X the
+same code runs over RCCL/xGMI)."""
+import multiprocessing
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from min_tfs_client_amd.parallel import shard_sizes
+
+
+def test_shard_sizes_even():
+    assert shard_sizes(256, 8) == [32] * 8
+
+
+def test_shard_sizes_uneven():
+    assert shard_sizes(10, 4) == [3, 3, 2, 2]
+    assert shard_sizes(3, 4) == [1, 1, 1, 0]
+
+
+def _dp_worker(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        import torch.distributed as dist
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+
+        import sys
+        root = os.path.dirname(os.path.dirname(
+            os.path.dirname(os.path.abspath(__file__))))
+        if root not in sys.path:
+            sys.path.insert(0, root)
+        from min_tfs_client_amd.parallel import DataParallelPredictor
+        from min_tfs_client_amd.server import ModelServer, identity_servable
+        from min_tfs_client_amd.turbo import TurboPredictClient
+
+        sock = f"unix:///tmp/dp_test_{os.getpid()}_{rank}.sock"
+        with ModelServer(address=sock, raw_predict=True) as srv:
+            srv.manager.load("m", identity_servable(), version=1)
+            with TurboPredictClient(sock) as client:
+                dp = DataParallelPredictor(client, device="cpu")
+                # uneven batch: 7 rows over 2 ranks
+                if rank == 0:
+                    torch.manual_seed(0)
+                    full = {"x": torch.randn(7, 3),
+                            "y": torch.arange(14,
+                                              dtype=torch.int64).view(7, 2)}
+                else:
+                    full = None
+                out = dp.predict("m", full)
+                # every rank must hold the full gathered batch
+                assert out["x"].shape == (7, 3)
+                assert out["y"].shape == (7, 2)
+                if rank == 0:
+                    assert torch.equal(out["x"], full["x"])
+                    assert torch.equal(out["y"], full["y"])
+                # shard-only mode
+                out_shard = dp.predict("m", full, gather_outputs=False)
+                assert out_shard["x"].shape[0] == shard_sizes(7, world)[rank]
+        dist.destroy_process_group()
+        q.put((rank, "ok"))
+    except Exception as e:  # noqa: BLE001
+        import traceback
+        q.put((rank, f"FAIL: {e}\n{traceback.format_exc()}"))
+
+
+@pytest.mark.timeout(120)
+def test_dp_predict_two_ranks():
+    import socket
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    ctx = multiprocessing.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_dp_worker, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=100) for _ in range(2)]
+    for p in procs:
+        p.join(timeout=30)
+    for rank, status in results:
+        assert status == "ok", f"rank {rank}: {status}"
