@@ -83,6 +83,16 @@ ModelServerConfig = _s.get_message_class("tensorflow.serving.ModelServerConfig")
 ReloadConfigRequest = _s.get_message_class("tensorflow.serving.ReloadConfigRequest")
 ReloadConfigResponse = _s.get_message_class(
     "tensorflow.serving.ReloadConfigResponse")
+PrometheusConfig = _s.get_message_class("tensorflow.serving.PrometheusConfig")
+SSLConfig = _s.get_message_class("tensorflow.serving.SSLConfig")
+MonitoringConfig = _s.get_message_class("tensorflow.serving.MonitoringConfig")
+LogMetadata = _s.get_message_class("tensorflow.serving.LogMetadata")
+ClassifyLog = _s.get_message_class("tensorflow.serving.ClassifyLog")
+RegressLog = _s.get_message_class("tensorflow.serving.RegressLog")
+PredictLog = _s.get_message_class("tensorflow.serving.PredictLog")
+MultiInferenceLog = _s.get_message_class(
+    "tensorflow.serving.MultiInferenceLog")
+PredictionLog = _s.get_message_class("tensorflow.serving.PredictionLog")
 
 
 class _EnumShim:

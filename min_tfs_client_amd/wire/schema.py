@@ -703,6 +703,90 @@ FILE_PREDICTION_SERVICE = proto_file(
     ])],
 )
 
+FILE_SSL_CONFIG = proto_file(
+    "tensorflow_serving/config/ssl_config.proto", "tensorflow.serving",
+    messages=[
+        message("SSLConfig",
+                fields=[field("server_key", 1, "string"),
+                        field("server_cert", 2, "string"),
+                        field("custom_ca", 3, "string"),
+                        field("client_verify", 4, "bool")]),
+    ],
+)
+
+FILE_MONITORING_CONFIG = proto_file(
+    "tensorflow_serving/config/monitoring_config.proto", "tensorflow.serving",
+    messages=[
+        message("PrometheusConfig", fields=[field("enable", 1, "bool"),
+                                            field("path", 2, "string")]),
+        message("MonitoringConfig",
+                fields=[field("prometheus_config", 1,
+                              ".tensorflow.serving.PrometheusConfig")]),
+    ],
+)
+
+FILE_CORE_LOGGING = proto_file(
+    "tensorflow_serving/core/logging.proto", "tensorflow.serving",
+    deps=["tensorflow_serving/apis/model.proto",
+          "tensorflow_serving/config/logging_config.proto"],
+    messages=[
+        message("LogMetadata",
+                fields=[field("model_spec", 1,
+                              ".tensorflow.serving.ModelSpec"),
+                        field("sampling_config", 2,
+                              ".tensorflow.serving.SamplingConfig"),
+                        field("saved_model_tags", 3, "string",
+                              repeated=True)]),
+    ],
+)
+
+# prediction_log.proto (warmup/record-log format). The session_run_log
+# oneof arm (field 5) is omitted — its SessionRunRequest closure pulls in
+# the whole RunOptions/debugger graph; unknown fields round-trip anyway.
+FILE_PREDICTION_LOG = proto_file(
+    "tensorflow_serving/apis/prediction_log.proto", "tensorflow.serving",
+    deps=["tensorflow_serving/apis/classification.proto",
+          "tensorflow_serving/apis/inference.proto",
+          "tensorflow_serving/apis/predict.proto",
+          "tensorflow_serving/apis/regression.proto",
+          "tensorflow_serving/core/logging.proto"],
+    messages=[
+        message("ClassifyLog",
+                fields=[field("request", 1,
+                              ".tensorflow.serving.ClassificationRequest"),
+                        field("response", 2,
+                              ".tensorflow.serving.ClassificationResponse")]),
+        message("RegressLog",
+                fields=[field("request", 1,
+                              ".tensorflow.serving.RegressionRequest"),
+                        field("response", 2,
+                              ".tensorflow.serving.RegressionResponse")]),
+        message("PredictLog",
+                fields=[field("request", 1,
+                              ".tensorflow.serving.PredictRequest"),
+                        field("response", 2,
+                              ".tensorflow.serving.PredictResponse")]),
+        message("MultiInferenceLog",
+                fields=[field("request", 1,
+                              ".tensorflow.serving.MultiInferenceRequest"),
+                        field("response", 2,
+                              ".tensorflow.serving.MultiInferenceResponse")]),
+        message("PredictionLog",
+                fields=[field("log_metadata", 1,
+                              ".tensorflow.serving.LogMetadata"),
+                        field("classify_log", 2,
+                              ".tensorflow.serving.ClassifyLog", oneof=0),
+                        field("regress_log", 3,
+                              ".tensorflow.serving.RegressLog", oneof=0),
+                        field("predict_log", 6,
+                              ".tensorflow.serving.PredictLog", oneof=0),
+                        field("multi_inference_log", 4,
+                              ".tensorflow.serving.MultiInferenceLog",
+                              oneof=0)],
+                oneofs=["log_type"]),
+    ],
+)
+
 # Dependency-ordered registration list.
 _ALL_FILES = [
     FILE_TYPES, FILE_TENSOR_SHAPE, FILE_RESOURCE_HANDLE, FILE_TENSOR,
@@ -712,6 +796,8 @@ _ALL_FILES = [
     FILE_GET_MODEL_STATUS, FILE_LOG_COLLECTOR_CONFIG, FILE_LOGGING_CONFIG,
     FILE_FS_STORAGE_PATH_SOURCE, FILE_MODEL_SERVER_CONFIG,
     FILE_MODEL_MANAGEMENT, FILE_MODEL_SERVICE, FILE_PREDICTION_SERVICE,
+    FILE_MONITORING_CONFIG, FILE_CORE_LOGGING, FILE_PREDICTION_LOG,
+    FILE_SSL_CONFIG,
 ]
 
 _pool = descriptor_pool.Default()
