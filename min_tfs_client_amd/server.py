@@ -195,11 +195,13 @@ def _abort(context, code, msg):
 class PredictionServiceImpl(PredictionServiceServicer):
     def __init__(self, manager: ModelManager,
                  output_encoding: str = "tensor_content",
-                 metrics: Optional[MetricsRegistry] = None):
+                 metrics: Optional[MetricsRegistry] = None,
+                 request_logger=None):
         assert output_encoding in ("tensor_content", "typed")
         self._manager = manager
         self._use_content = output_encoding == "tensor_content"
         self.metrics = metrics or MetricsRegistry()
+        self.request_logger = request_logger
 
     # -- helpers --------------------------------------------------------
     def _resolve(self, model_spec, context):
@@ -242,6 +244,9 @@ class PredictionServiceImpl(PredictionServiceServicer):
             request.model_spec.signature_name or "serving_default")
         self._encode_outputs(response, outputs, request.output_filter)
         self.metrics.observe_request("predict", time.perf_counter() - t0)
+        if self.request_logger is not None:
+            self.request_logger.log_predict(request.model_spec.name,
+                                            request, response)
         return response
 
     def Classify(self, request, context):
@@ -393,7 +398,7 @@ class ModelServiceImpl(ModelServiceServicer):
 # ---------------------------------------------------------------------------
 
 def _raw_predict_handler(manager: ModelManager, device: str,
-                         metrics: MetricsRegistry):
+                         metrics: MetricsRegistry, request_logger=None):
     """Raw-bytes Predict: request bytes in, response bytes out, all codec
     work in the C++ extension (turbo path — see turbo.py)."""
     from .ops import require_native
@@ -415,6 +420,9 @@ def _raw_predict_handler(manager: ModelManager, device: str,
         if getattr(servable, "is_identity", False):
             out = native.echo_predict(data)
             metrics.observe_request("predict", time.perf_counter() - t0)
+            if request_logger is not None:
+                request_logger.log_predict(spec["name"], bytes(data),
+                                           bytes(out))
             return out
         try:
             outputs = servable(inputs)
@@ -431,6 +439,9 @@ def _raw_predict_handler(manager: ModelManager, device: str,
             spec["name"], -1 if version is None else version,
             spec["signature_name"] or "serving_default", names, tensors, 0)
         metrics.observe_request("predict", time.perf_counter() - t0)
+        if request_logger is not None:
+            request_logger.log_predict(spec["name"], bytes(data),
+                                       bytes(blob))
         return blob
 
     def identity(x):
@@ -460,15 +471,18 @@ class ModelServer:
                  address: Optional[str] = None):
         from .utils.allocator import tune_malloc
         tune_malloc()
+        from .request_logging import ServerRequestLogger
         self.manager = manager or ModelManager()
         self.metrics = MetricsRegistry()
+        self.request_logger = ServerRequestLogger()
         self._server = grpc.server(
             futures.ThreadPoolExecutor(max_workers=max_workers),
             options=[("grpc.max_send_message_length", 1 << 30),
                      ("grpc.max_receive_message_length", 1 << 30),
                      ("grpc.http2.max_frame_size", 16 * 1024 * 1024 - 1)])
         self.prediction_service = PredictionServiceImpl(
-            self.manager, output_encoding, self.metrics)
+            self.manager, output_encoding, self.metrics,
+            self.request_logger)
         self.model_service = ModelServiceImpl(self.manager, servable_factory)
         if raw_predict:
             from .wire import messages as _pb
@@ -476,7 +490,8 @@ class ModelServer:
                 add_ModelServiceServicer_to_server as _add_ms)
             handlers = {
                 "Predict": _raw_predict_handler(self.manager, device,
-                                                self.metrics),
+                                                self.metrics,
+                                                self.request_logger),
                 "Classify": grpc.unary_unary_rpc_method_handler(
                     self.prediction_service.Classify,
                     request_deserializer=_pb.ClassificationRequest.FromString,
