@@ -138,7 +138,8 @@ def main():
 
     if args.encoding == "turbo":
         from min_tfs_client_amd.turbo import TurboPredictClient
-        client = TurboPredictClient(address)
+        client = TurboPredictClient(
+            address, num_channels=min(args.pipeline, 4))
 
         def step_fn(step_inputs):
             out_dev = f"cuda:{local_rank}" if has_gpu else "cpu"

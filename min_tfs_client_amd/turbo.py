@@ -53,22 +53,39 @@ class TurboPredictClient:
 
     def __init__(self, target: str,
                  credentials: Optional[grpc.ChannelCredentials] = None,
-                 options: Optional[list] = None):
+                 options: Optional[list] = None,
+                 num_channels: int = 1):
         self._native = require_native()
         tune_malloc()  # large wire buffers: arena reuse, no per-call mmap
         opts = _CHANNEL_OPTS + (options or [])
-        if credentials:
-            self._channel = grpc.secure_channel(target, credentials,
-                                                options=opts)
-        else:
-            self._channel = grpc.insecure_channel(target, options=opts)
-        self._predict = self._channel.unary_unary(
-            _PREDICT_PATH, request_serializer=_identity,
-            response_deserializer=_identity)
+        self._channels = []
+        self._stubs = []
+        for i in range(max(1, num_channels)):
+            # separate HTTP/2 connections (no shared subchannel) ->
+            # parallel transport for pipelined requests
+            copts = opts + [("grpc.use_local_subchannel_pool", 1),
+                            ("grpc.channel_id", i)]
+            if credentials:
+                ch = grpc.secure_channel(target, credentials, options=copts)
+            else:
+                ch = grpc.insecure_channel(target, options=copts)
+            self._channels.append(ch)
+            self._stubs.append(ch.unary_unary(
+                _PREDICT_PATH, request_serializer=_identity,
+                response_deserializer=_identity))
+        self._channel = self._channels[0]
+        self._predict = self._stubs[0]
+        self._rr = 0
         self.metrics = MetricsRegistry()
 
+    def _next_stub(self):
+        stub = self._stubs[self._rr % len(self._stubs)]
+        self._rr += 1
+        return stub
+
     def close(self):
-        self._channel.close()
+        for ch in self._channels:
+            ch.close()
 
     def __enter__(self):
         return self
@@ -112,7 +129,7 @@ class TurboPredictClient:
         decode) — call decode(future.result()) to get output tensors."""
         blob = self.serialize_request(model_name, inputs, model_version,
                                       signature_name, copy_mode)
-        fut = self._predict.future(blob, timeout)
+        fut = self._next_stub().future(blob, timeout)
 
         def decode(resp_bytes, output_device="cpu"):
             _s, outputs, _ = self._native.parse_predict_response(
