@@ -161,19 +161,29 @@ def main():
                     for k, v in resp.outputs.items()}
 
     # ---- one data-parallel step --------------------------------------
-    # world_size > 1 (config 4): rank0's global batch is scattered over
-    # RCCL, each rank round-trips its shard, responses all-gathered.
+    # world_size > 1 (config 4): rank 0 holds the GLOBAL batch
+    # (per_rank_batch * N rows); each step scatters the real dim-0 chunks
+    # over RCCL/xGMI, each rank round-trips its shard, responses are
+    # all-gathered.
     keys = sorted(inputs.keys())
+    if dist is not None and rank == 0:
+        global_inputs = {
+            k: torch.cat([inputs[k]] * world_size, dim=0).contiguous()
+            if inputs[k].dim() > 0 else inputs[k]
+            for k in keys}
+    else:
+        global_inputs = None
 
     def dp_step():
         step_inputs = {}
         if dist is not None:
             for k in keys:
-                full = inputs[k]
-                shard = torch.empty_like(full)
+                shard = torch.empty_like(inputs[k])
                 if rank == 0:
-                    scatter_list = [full.contiguous()
-                                    for _ in range(world_size)]
+                    full = global_inputs[k]
+                    n = inputs[k].shape[0] if inputs[k].dim() > 0 else 1
+                    scatter_list = list(full.split(n, dim=0)) \
+                        if full.dim() > 0 else [full] * world_size
                 else:
                     scatter_list = None
                 dist.scatter(shard, scatter_list, src=0)

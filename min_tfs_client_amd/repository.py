@@ -282,23 +282,25 @@ class FileSystemStoragePathSource:
             if not found and self.fail_if_zero:
                 raise RuntimeError(
                     f"model {name}: no versions at startup under {base}")
-            aspired = set(policy.aspired(found))
+            aspired = set(policy.aspired(sorted(found)))
             loaded = self._loaded.setdefault(name, {})
             for ver in sorted(aspired - set(loaded)):
-                vdir = os.path.join(base, str(ver))
+                vdir = os.path.join(base, found[ver])
                 if self._load_with_retries(name, ver, vdir):
                     loaded[ver] = vdir
             for ver in sorted(set(loaded) - aspired):
                 self.manager.unload(name, ver)
                 del loaded[ver]
 
-    def _scan_versions(self, base: str) -> List[int]:
+    def _scan_versions(self, base: str) -> Dict[int, str]:
+        """{version int: dir entry name} — zero-padded names like the
+        reference fixture's 00000001 resolve to version 1."""
         if not os.path.isdir(base):
             raise FileNotFoundError(base)
-        out = []
+        out: Dict[int, str] = {}
         for entry in os.listdir(base):
             if entry.isdigit() and os.path.isdir(os.path.join(base, entry)):
-                out.append(int(entry))
+                out[int(entry)] = entry
         return out
 
     def _load_with_retries(self, name: str, version: int,
