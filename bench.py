@@ -37,13 +37,24 @@ import numpy as np  # noqa: E402
 import torch  # noqa: E402
 
 
-def _server_proc(address: str, ready, stop):
+def _server_proc(address: str, ready, stop, servable: str = "echo",
+                 device: str = "cpu"):
     """Loopback PredictionService in its own process (own GIL)."""
     if _ROOT not in sys.path:
         sys.path.insert(0, _ROOT)
     from min_tfs_client_amd.server import ModelServer, identity_servable
-    with ModelServer(address=address, raw_predict=True, max_workers=8) as srv:
-        srv.manager.load("default", identity_servable(), version=1)
+    with ModelServer(address=address, raw_predict=True, max_workers=8,
+                     device=device) as srv:
+        if servable == "echo":
+            srv.manager.load("default", identity_servable(), version=1)
+        elif servable == "resnet50":
+            from min_tfs_client_amd.models import resnet50_servable
+            srv.manager.load("default", resnet50_servable(device), version=1)
+        elif servable == "bert":
+            from min_tfs_client_amd.models import bert_servable
+            srv.manager.load("default", bert_servable(device), version=1)
+        else:
+            raise ValueError(f"unknown servable {servable}")
         ready.set()
         stop.wait()
 
@@ -85,6 +96,10 @@ def main():
                     help="0 = pinned-staged pipelined copies, 1 = direct "
                          "pageable hipMemcpy (A/B)")
     ap.add_argument("--transport", default="unix", choices=["unix", "tcp"])
+    ap.add_argument("--servable", default="echo",
+                    choices=["echo", "resnet50", "bert"],
+                    help="what the loopback server runs: echo (the codec/"
+                         "transport benchmark) or a real model family")
     ap.add_argument("--pipeline", type=int, default=1,
                     help="in-flight requests per rank (1 = sequential; "
                          ">1 overlaps serialize/transport/parse of "
@@ -125,7 +140,10 @@ def main():
         port = s.getsockname()[1]
         s.close()
         address = f"127.0.0.1:{port}"
-    proc = ctx.Process(target=_server_proc, args=(address, ready, stop),
+    server_device = f"cuda:{local_rank}" if has_gpu else "cpu"
+    proc = ctx.Process(target=_server_proc,
+                       args=(address, ready, stop, args.servable,
+                             server_device),
                        daemon=True)
     proc.start()
     if not ready.wait(60):
@@ -284,7 +302,9 @@ def main():
                       else "fp32"),
             "data": "synthetic",
             "config": {
-                "model": f"identity-echo[{args.bench_config}]",
+                "model": (f"identity-echo[{args.bench_config}]"
+                          if args.servable == "echo"
+                          else f"{args.servable}[{args.bench_config}]"),
                 "global_batch": per_rank_batch * n_gpus,
                 "seq_len": 512 if args.bench_config == "bert" else None,
                 "shape_per_request": list(
