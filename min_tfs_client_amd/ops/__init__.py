@@ -58,6 +58,13 @@ def nchw_to_nhwc(tensor, out_dtype=None):
     return require_native().nchw_to_nhwc(tensor, out_dtype)
 
 
+def nhwc_to_nchw(tensor, out_dtype=None):
+    """Inverse fused layout transform (unpack direction)."""
+    if out_dtype is None:
+        out_dtype = tensor.dtype
+    return require_native().nhwc_to_nchw(tensor, out_dtype)
+
+
 def quantize_q8(tensor, scale, zero_point=0.0):
     return require_native().quantize_q8(tensor, scale, zero_point)
 

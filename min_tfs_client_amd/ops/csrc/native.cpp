@@ -21,6 +21,7 @@ namespace py = pybind11;
 namespace mi355x {
 at::Tensor cast_op(const at::Tensor& in, at::ScalarType out_dtype);
 at::Tensor nchw_to_nhwc(const at::Tensor& in, at::ScalarType out_dtype);
+at::Tensor nhwc_to_nchw(const at::Tensor& in, at::ScalarType out_dtype);
 at::Tensor quantize_q8(const at::Tensor& in, double scale, double zp);
 at::Tensor dequantize_q8(const at::Tensor& in, double scale, double zp);
 void copy_device_to_host_ptr(const at::Tensor& src, void* dst, size_t n,
@@ -322,6 +323,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   // HIP ops
   m.def("cast", &mi355x::cast_op, py::arg("input"), py::arg("out_dtype"));
   m.def("nchw_to_nhwc", &mi355x::nchw_to_nhwc, py::arg("input"),
+        py::arg("out_dtype"));
+  m.def("nhwc_to_nchw", &mi355x::nhwc_to_nchw, py::arg("input"),
         py::arg("out_dtype"));
   m.def("quantize_q8", &mi355x::quantize_q8, py::arg("input"),
         py::arg("scale"), py::arg("zero_point") = 0.0);

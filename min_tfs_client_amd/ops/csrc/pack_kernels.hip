@@ -210,6 +210,58 @@ __global__ void nchw_nhwc_smallc_kernel(
   }
 }
 
+// Inverse direction (NHWC -> NCHW), small compile-time C: each lane reads
+// HWPL*C contiguous inputs (vector loads) and writes HWPL consecutive
+// elements into each of C channel streams (vector stores per channel).
+template <typename CVT, int C, int HWPL>
+__global__ void nhwc_nchw_smallc_kernel(
+    const typename CVT::In* __restrict__ in,
+    typename CVT::Out* __restrict__ out,
+    int64_t HW, int64_t N) {
+  using In = typename CVT::In;
+  using Out = typename CVT::Out;
+  const int64_t hw_chunks = (HW + HWPL - 1) / HWPL;
+  int64_t idx0 = int64_t(blockIdx.x) * blockDim.x + threadIdx.x;
+  int64_t stride = int64_t(gridDim.x) * blockDim.x;
+  for (int64_t idx = idx0; idx < N * hw_chunks; idx += stride) {
+    const int64_t nimg = idx / hw_chunks;
+    const int64_t hw0 = (idx - nimg * hw_chunks) * HWPL;
+    const In* src = in + nimg * HW * C + hw0 * C;
+    Out* dst0 = out + nimg * C * HW + hw0;
+    if (hw0 + HWPL <= HW &&
+        (sizeof(In) * HWPL * C) % 16 == 0 &&
+        (reinterpret_cast<uintptr_t>(src) & 15) == 0 &&
+        (sizeof(Out) * HWPL) % 16 == 0 &&
+        (reinterpret_cast<uintptr_t>(dst0) & 15) == 0 &&
+        (HW * sizeof(Out)) % 16 == 0) {
+      In vin[HWPL * C];
+      constexpr int NLV = int(sizeof(In) * HWPL * C / 16);
+#pragma unroll
+      for (int v = 0; v < NLV; ++v)
+        reinterpret_cast<int4*>(vin)[v] =
+            reinterpret_cast<const int4*>(src)[v];
+      Out vout[C][HWPL];
+#pragma unroll
+      for (int c = 0; c < C; ++c)
+#pragma unroll
+        for (int k = 0; k < HWPL; ++k)
+          vout[c][k] = CVT::cvt(vin[k * C + c]);
+      constexpr int NSV = int(sizeof(Out) * HWPL / 16);
+#pragma unroll
+      for (int c = 0; c < C; ++c)
+#pragma unroll
+        for (int v = 0; v < NSV; ++v)
+          reinterpret_cast<int4*>(dst0 + c * HW)[v] =
+              *reinterpret_cast<const int4*>(&vout[c][v * (16 / sizeof(Out))]);
+    } else {
+      const int span = int(min(int64_t(HWPL), HW - hw0));
+      for (int k = 0; k < span; ++k)
+        for (int c = 0; c < C; ++c)
+          dst0[c * HW + k] = CVT::cvt(src[k * C + c]);
+    }
+  }
+}
+
 // Runtime-C fallback for 5..8 channels (4 hw per lane).
 template <typename CVT, int MAXC>
 __global__ void nchw_nhwc_midc_kernel(
@@ -467,6 +519,77 @@ at::Tensor nchw_to_nhwc(const at::Tensor& in, at::ScalarType out_dtype) {
     launch_nchw_nhwc<F32ToBf16>(in, out, N, C, HW);
   else
     TORCH_CHECK(false, "nchw_to_nhwc: unsupported dtype pair");
+  return out;
+}
+
+// Inverse layout transform: NHWC -> NCHW with optional cast (the unpack
+// direction: responses arrive NHWC fp32, models want NCHW bf16).
+at::Tensor nhwc_to_nchw(const at::Tensor& in, at::ScalarType out_dtype) {
+  TORCH_CHECK(in.is_cuda() && in.dim() == 4 && in.is_contiguous(),
+              "nhwc_to_nchw expects a contiguous 4-D device tensor [N,H,W,C]");
+  const int64_t N = in.size(0), H = in.size(1), W = in.size(2),
+                C = in.size(3);
+  const int64_t HW = H * W;
+  auto out = at::empty({N, C, H, W}, in.options().dtype(out_dtype));
+  auto id = in.scalar_type();
+
+  auto launch_small = [&](auto cvt_tag) {
+    using CVT = decltype(cvt_tag);
+    const int block = 256;
+    const int64_t hw_chunks = (HW + 7) / 8;
+    const int grid = grid_for(N * hw_chunks, block);
+    auto* src = reinterpret_cast<const typename CVT::In*>(
+        in.const_data_ptr());
+    auto* dst = reinterpret_cast<typename CVT::Out*>(
+        out.mutable_data_ptr());
+    switch (C) {
+      case 1:
+        hipLaunchKernelGGL((nhwc_nchw_smallc_kernel<CVT, 1, 8>), dim3(grid),
+                           dim3(block), 0, current_stream(), src, dst, HW, N);
+        break;
+      case 2:
+        hipLaunchKernelGGL((nhwc_nchw_smallc_kernel<CVT, 2, 8>), dim3(grid),
+                           dim3(block), 0, current_stream(), src, dst, HW, N);
+        break;
+      case 3:
+        hipLaunchKernelGGL((nhwc_nchw_smallc_kernel<CVT, 3, 8>), dim3(grid),
+                           dim3(block), 0, current_stream(), src, dst, HW, N);
+        break;
+      default:
+        hipLaunchKernelGGL((nhwc_nchw_smallc_kernel<CVT, 4, 8>), dim3(grid),
+                           dim3(block), 0, current_stream(), src, dst, HW, N);
+    }
+  };
+  auto launch_tiled = [&](auto cvt_tag) {
+    using CVT = decltype(cvt_tag);
+    // generic batched 2-D transpose: input rows = HW, cols = C
+    constexpr int TILE = 64;
+    dim3 grid(unsigned((C + TILE - 1) / TILE),
+              unsigned((HW + TILE - 1) / TILE), unsigned(N));
+    hipLaunchKernelGGL(nchw_nhwc_tiled_kernel<CVT>, grid, dim3(256), 0,
+                       current_stream(),
+                       reinterpret_cast<const typename CVT::In*>(
+                           in.const_data_ptr()),
+                       reinterpret_cast<typename CVT::Out*>(
+                           out.mutable_data_ptr()),
+                       /*C=*/HW, /*HW=*/C, N);
+  };
+
+  auto dispatch = [&](auto cvt_tag) {
+    if (C <= 4) launch_small(cvt_tag);
+    else launch_tiled(cvt_tag);
+  };
+  if (id == at::kFloat && out_dtype == at::kBFloat16)
+    dispatch(F32ToBf16{});
+  else if (id == at::kFloat && out_dtype == at::kFloat)
+    dispatch(IdF32{});
+  else if (id == at::kBFloat16 && out_dtype == at::kFloat)
+    dispatch(Bf16ToF32{});
+  else if (id == at::kHalf && out_dtype == at::kFloat)
+    dispatch(F16ToF32{});
+  else
+    TORCH_CHECK(false, "nhwc_to_nchw: unsupported dtype pair");
+  HIP_CHECK(hipGetLastError());
   return out;
 }
 

@@ -135,3 +135,28 @@ def test_serialize_mixed_cpu_gpu_inputs(native):
     _, outs, _ = native.parse_predict_request(blob, "cpu", 0)
     assert torch.equal(outs["a"], a.cpu())
     assert torch.equal(outs["b"], b)
+
+
+# ---------------------------------------------------------------------------
+# inverse layout transform (unpack direction)
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("shape", [
+    (32, 224, 224, 3),      # NHWC image (small-C path)
+    (4, 17, 31, 3),
+    (2, 56, 56, 256),       # generic tiled path
+    (3, 7, 9, 65),
+])
+def test_nhwc_nchw_f32_to_bf16(native, shape):
+    x = torch.randn(*shape, device=DEV, dtype=torch.float32)
+    out = ops.nhwc_to_nchw(x, torch.bfloat16)
+    ref = x.permute(0, 3, 1, 2).contiguous().to(torch.bfloat16)
+    assert out.shape == ref.shape
+    assert torch.equal(out, ref)
+
+
+def test_nhwc_nchw_roundtrip_with_forward(native):
+    x = torch.randn(8, 3, 64, 64, device=DEV, dtype=torch.bfloat16)
+    nhwc = ops.nchw_to_nhwc(x, torch.float32)
+    back = ops.nhwc_to_nchw(nhwc, torch.bfloat16)
+    assert torch.equal(back, x)
