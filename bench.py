@@ -146,7 +146,8 @@ def main():
                              server_device),
                        daemon=True)
     proc.start()
-    if not ready.wait(60):
+    # fresh boxes: first torch import in the child can take minutes
+    if not ready.wait(300):
         raise RuntimeError("bench server failed to start")
 
     # ---- client ------------------------------------------------------

@@ -28,6 +28,7 @@ except ImportError:  # pragma: no cover
 from .ops import require_native
 from .utils.allocator import tune_malloc
 from .utils.metrics import MetricsRegistry
+from .utils.tracing import trace_span
 
 _PREDICT_PATH = "/tensorflow.serving.PredictionService/Predict"
 
@@ -114,12 +115,17 @@ class TurboPredictClient:
         """One Predict round trip. ``output_device``: where response
         tensors land ("cpu" default; "cuda:N" unpacks over the staging
         pipeline straight to HBM)."""
-        blob = self.serialize_request(model_name, inputs, model_version,
-                                      signature_name, copy_mode)
-        resp = self._predict(blob, timeout)
+        with trace_span("turbo.serialize", model=model_name,
+                        bytes=sum(t.numel() * t.element_size()
+                                  for t in inputs.values())):
+            blob = self.serialize_request(model_name, inputs, model_version,
+                                          signature_name, copy_mode)
+        with trace_span("turbo.rpc", bytes=len(blob)):
+            resp = self._predict(blob, timeout)
         dev = str(output_device) if output_device is not None else "cpu"
-        _spec, outputs, _ = self._native.parse_predict_response(
-            resp, dev, copy_mode)
+        with trace_span("turbo.parse", device=dev):
+            _spec, outputs, _ = self._native.parse_predict_response(
+                resp, dev, copy_mode)
         return outputs
 
     def predict_future(self, model_name, inputs, timeout=60.0,

@@ -214,3 +214,40 @@ def test_logging_through_raw_server(tmp_path):
     log = pb.PredictionLog.FromString(records[0])
     assert "x" in log.predict_log.request.inputs
     assert "x" in log.predict_log.response.outputs
+
+
+# -- tracing -----------------------------------------------------------------
+
+def test_tracer_spans_and_export(tmp_path):
+    import torch
+    from min_tfs_client_amd.utils.tracing import Tracer
+    from min_tfs_client_amd.server import ModelServer, identity_servable
+    from min_tfs_client_amd.turbo import TurboPredictClient
+
+    tracer = Tracer.get()
+    tracer.clear()
+    tracer.start()
+    try:
+        with ModelServer(port=0, raw_predict=True) as srv:
+            srv.manager.load("m", identity_servable(), version=1)
+            with TurboPredictClient(srv.address) as c:
+                c.predict("m", {"x": torch.ones(4)})
+    finally:
+        tracer.stop()
+    out = str(tmp_path / "trace.json")
+    n = tracer.export(out)
+    assert n >= 3  # serialize, rpc, parse
+    import json as _json
+    events = _json.load(open(out))["traceEvents"]
+    names = {e["name"] for e in events}
+    assert {"turbo.serialize", "turbo.rpc", "turbo.parse"} <= names
+
+
+def test_tracer_disabled_is_noop():
+    from min_tfs_client_amd.utils.tracing import Tracer, trace_span
+    t = Tracer.get()
+    t.clear()
+    assert not t.enabled
+    with trace_span("nothing"):
+        pass
+    assert t.export("/dev/null") == 0
