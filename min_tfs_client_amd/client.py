@@ -34,14 +34,27 @@ TensorLike = Union[np.ndarray, "torch.Tensor"]
 class TensorServingClient:
     """gRPC client for a TF-Serving-compatible PredictionService."""
 
+    RETRY_SERVICE_CONFIG = {'methodConfig': [{'name': [{'service': 'tensorflow.serving.PredictionService'}, {'service': 'tensorflow.serving.ModelService'}], 'retryPolicy': {'maxAttempts': 4, 'initialBackoff': '0.05s', 'maxBackoff': '1s', 'backoffMultiplier': 2, 'retryableStatusCodes': ['UNAVAILABLE']}}]}
+
     def __init__(self, host: str, port: int,
                  credentials: Optional[grpc.ChannelCredentials] = None,
-                 options: Optional[list] = None) -> None:
+                 options: Optional[list] = None,
+                 enable_retries: bool = False) -> None:
         self._host_address = f"{host}:{port}"
         default_options = [
             ("grpc.max_send_message_length", 1 << 30),
             ("grpc.max_receive_message_length", 1 << 30),
         ]
+        if enable_retries:
+            # transparent UNAVAILABLE retries with backoff — the reference
+            # client has no retry story (SURVEY §5 failure detection);
+            # opt-in here so default behavior stays reference-identical
+            import json as _json
+            default_options += [
+                ("grpc.enable_retries", 1),
+                ("grpc.service_config",
+                 _json.dumps(self.RETRY_SERVICE_CONFIG)),
+            ]
         opts = default_options + (options or [])
         if credentials:
             self._channel = grpc.secure_channel(

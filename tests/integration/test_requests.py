@@ -165,3 +165,16 @@ def test_bf16_tensor_input(client):
     assert out_proto.dtype == 14
     from min_tfs_client_amd.tensors import tensor_proto_to_tensor
     assert torch.equal(tensor_proto_to_tensor(out_proto), t)
+
+
+def test_enable_retries_option(server):
+    """Retry-enabled channel still works (transparent retries on
+    UNAVAILABLE; opt-in — SURVEY §5 notes the reference has none)."""
+    c = TensorServingClient("127.0.0.1", server.port, enable_retries=True)
+    try:
+        x = np.ones(3, np.float32)
+        resp = c.predict_request("default", {"x": x})
+        np.testing.assert_array_equal(
+            tensor_proto_to_ndarray(resp.outputs["x"]), x)
+    finally:
+        c.close()
