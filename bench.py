@@ -92,7 +92,7 @@ def main():
                     choices=["turbo", "proto"],
                     help="turbo = C++ codec raw-bytes path; proto = "
                          "python-protobuf client (reference-style)")
-    ap.add_argument("--copy-mode", type=int, default=0,
+    ap.add_argument("--copy-mode", type=int, default=1,
                     help="0 = pinned-staged pipelined copies, 1 = direct "
                          "pageable hipMemcpy (A/B)")
     ap.add_argument("--transport", default="unix", choices=["unix", "tcp"])

@@ -89,5 +89,5 @@ def pack_tensor_proto(tensor):
     proto.dtype = dtype.enum
     for d in tensor.shape:
         proto.tensor_shape.dim.add().size = d
-    proto.tensor_content = n.tensor_content_bytes(tensor, 0)
+    proto.tensor_content = n.tensor_content_bytes(tensor, 1)
     return proto

@@ -99,7 +99,7 @@ class TurboPredictClient:
                           inputs: Dict[str, "torch.Tensor"],
                           model_version: Optional[int] = None,
                           signature_name: str = "",
-                          copy_mode: int = 0) -> bytes:
+                          copy_mode: int = 1) -> bytes:
         names = list(inputs.keys())
         tensors = [inputs[k] for k in names]
         return self._native.serialize_predict_request(
@@ -111,7 +111,7 @@ class TurboPredictClient:
                 model_version: Optional[int] = None,
                 signature_name: str = "",
                 output_device: Optional[Union[str, "torch.device"]] = None,
-                copy_mode: int = 0) -> Dict[str, "torch.Tensor"]:
+                copy_mode: int = 1) -> Dict[str, "torch.Tensor"]:
         """One Predict round trip. ``output_device``: where response
         tensors land ("cpu" default; "cuda:N" unpacks over the staging
         pipeline straight to HBM)."""
@@ -130,7 +130,7 @@ class TurboPredictClient:
 
     def predict_future(self, model_name, inputs, timeout=60.0,
                        model_version=None, signature_name="",
-                       copy_mode: int = 0):
+                       copy_mode: int = 1):
         """Async variant for request pipelining: returns (grpc future,
         decode) — call decode(future.result()) to get output tensors."""
         blob = self.serialize_request(model_name, inputs, model_version,
