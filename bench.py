@@ -43,7 +43,7 @@ def _server_proc(address: str, ready, stop, servable: str = "echo",
     if _ROOT not in sys.path:
         sys.path.insert(0, _ROOT)
     from min_tfs_client_amd.server import ModelServer, identity_servable
-    with ModelServer(address=address, raw_predict=True, max_workers=8,
+    with ModelServer(address=address, raw_predict=True, max_workers=16,
                      device=device) as srv:
         if servable == "echo":
             srv.manager.load("default", identity_servable(), version=1)
@@ -158,7 +158,7 @@ def main():
     if args.encoding == "turbo":
         from min_tfs_client_amd.turbo import TurboPredictClient
         client = TurboPredictClient(
-            address, num_channels=min(args.pipeline, 4))
+            address, num_channels=min(args.pipeline, 8))
 
         def step_fn(step_inputs):
             out_dev = f"cuda:{local_rank}" if has_gpu else "cpu"

@@ -107,7 +107,7 @@ def bert_servable(device: str = "cpu", dtype=torch.float32, seed: int = 0):
         return {"last_hidden_state": hidden.float(),
                 "pooled_output": pooled.float()}
 
-    return Servable(
+    s = Servable(
         fn,
         signature={
             "method_name": "tensorflow/serving/predict",
@@ -116,3 +116,5 @@ def bert_servable(device: str = "cpu", dtype=torch.float32, seed: int = 0):
             "outputs": {"last_hidden_state": (1, [-1, -1, 768]),
                         "pooled_output": (1, [-1, 768])},
         })
+    s.module = model
+    return s

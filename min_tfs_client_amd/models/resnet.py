@@ -90,10 +90,12 @@ def resnet50_servable(device: str = "cpu", dtype=torch.float32,
         x = x.to(device=device, dtype=dtype)
         return {"logits": model(x).float()}
 
-    return Servable(
+    s = Servable(
         fn,
         signature={
             "method_name": "tensorflow/serving/predict",
             "inputs": {"images": (1, [-1, 3, 224, 224])},   # DT_FLOAT
             "outputs": {"logits": (1, [-1, 1000])},
         })
+    s.module = model
+    return s

@@ -169,11 +169,14 @@ def default_loader(name: str, version_dir: str) -> Servable:
         state = cfg.get("state_dict")
         if state:
             import torch
+            module = getattr(servable, "module", None)
+            if module is None:
+                raise ValueError(
+                    f"model family {family!r} does not expose a module for "
+                    f"state_dict loading")
             sd = torch.load(os.path.join(version_dir, state),
                             map_location=device, weights_only=True)
-            # servables hold their module in the closure; expose via attr
-            raise ValueError("state_dict loading requires a module-bearing "
-                             "servable; use model.pt TorchScript instead")
+            module.load_state_dict(sd)
         return servable
     pt_path = os.path.join(version_dir, "model.pt")
     if os.path.exists(pt_path):

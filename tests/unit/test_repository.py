@@ -230,3 +230,22 @@ def test_warmup_disabled(tmp_path):
     src.set_models({"m": str(tmp_path / "m")})
     src.poll_once()
     assert mgr.version_statuses("m")[0][1] == 30
+
+
+def test_loader_state_dict(tmp_path):
+    """model.json with a state_dict: weights are loaded into the family
+    module."""
+    import json as _json
+    vdir = tmp_path / "m" / "1"
+    vdir.mkdir(parents=True)
+    from min_tfs_client_amd.models import resnet50_servable
+    donor = resnet50_servable()
+    torch.save(donor.module.state_dict(), str(vdir / "weights.pt"))
+    (vdir / "model.json").write_text(_json.dumps(
+        {"family": "resnet50", "state_dict": "weights.pt"}))
+    s = default_loader("m", str(vdir))
+    # weights equal the donor's, not a fresh random init
+    for (n1, p1), (n2, p2) in zip(s.module.state_dict().items(),
+                                  donor.module.state_dict().items()):
+        assert n1 == n2
+        assert torch.equal(p1, p2)
