@@ -646,6 +646,11 @@ class StagingPool {
 
   static constexpr size_t kChunk = 8u << 20;  // 8 MiB per staging buffer
 
+  // d2h/h2d use one stream + two staging buffers: serialize concurrent
+  // callers (gRPC worker threads) behind a mutex. Measured: concurrent
+  // unsynchronized use corrupts staging hand-offs and stalls on events.
+  std::mutex mu_;
+
   void ensure_init() {
     std::call_once(init_flag_, [this] {
       HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
@@ -661,6 +666,7 @@ class StagingPool {
   // device -> host (dst pageable), pipelined through pinned staging
   void d2h(void* dst, const void* src_dev, size_t nbytes) {
     ensure_init();
+    std::lock_guard<std::mutex> lock(mu_);
     size_t nchunks = (nbytes + kChunk - 1) / kChunk;
     // issue chunk 0
     size_t issued = 0;
@@ -693,6 +699,7 @@ class StagingPool {
   // host (pageable src) -> device, pipelined through pinned staging
   void h2d(void* dst_dev, const void* src, size_t nbytes) {
     ensure_init();
+    std::lock_guard<std::mutex> lock(mu_);
     size_t nchunks = (nbytes + kChunk - 1) / kChunk;
     for (size_t c = 0; c < nchunks; ++c) {
       size_t off = c * kChunk;

@@ -408,7 +408,7 @@ def _raw_predict_handler(manager: ModelManager, device: str,
         native = require_native()
         try:
             spec, inputs, _filter = native.parse_predict_request(
-                data, device, 0)
+                data, device, 1)
         except Exception as e:  # noqa: BLE001
             _abort(context, grpc.StatusCode.INVALID_ARGUMENT,
                    f"request parsing error: {e}")
@@ -437,7 +437,7 @@ def _raw_predict_handler(manager: ModelManager, device: str,
             tensors.append(v)
         blob = native.serialize_predict_response(
             spec["name"], -1 if version is None else version,
-            spec["signature_name"] or "serving_default", names, tensors, 0)
+            spec["signature_name"] or "serving_default", names, tensors, 1)
         metrics.observe_request("predict", time.perf_counter() - t0)
         if request_logger is not None:
             request_logger.log_predict(spec["name"], bytes(data),
