@@ -58,16 +58,26 @@ def main():
     with ModelServer(address=sock, raw_predict=True, device=device,
                      max_workers=32) as srv:
         srv.manager.load("plain", resnet50_servable(device), version=1)
+        # allowed_batch_sizes pins the merged shapes so MIOpen's per-shape
+        # kernel search happens once per size, not per request mix (the
+        # reason TF-Serving has the same knob, batching_session.h:92-97)
         srv.manager.load(
             "batched",
             BatchingServable(resnet50_servable(device), max_batch_size=64,
-                             batch_timeout_s=0.003),
+                             batch_timeout_s=0.003,
+                             allowed_batch_sizes=[8, 16, 32, 64]),
             version=1)
         n_threads, per_thread, batch = 8, 12, 4
         if device == "cpu":
             n_threads, per_thread = 2, 3
+        # shape warmup: one request per allowed batch size
+        with TurboPredictClient(sock) as wc:
+            for b in (4, 8, 16, 32, 64):
+                x = torch.randn(b, 3, 224, 224, device=device)
+                wc.predict("plain", {"images": x}, timeout=300)
+                wc.predict("batched", {"images": x}, timeout=300)
         for model in ("plain", "batched"):
-            # warmup
+            # concurrency warmup
             run_load(sock, model, 2, 2, batch, device)
             r = run_load(sock, model, n_threads, per_thread, batch, device)
             r.update({"model": model, "threads": n_threads,
