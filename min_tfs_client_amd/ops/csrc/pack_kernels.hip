@@ -343,23 +343,27 @@ __global__ void nchw_nhwc_tiled_kernel(
 // int8 affine quantize / dequantize
 // ---------------------------------------------------------------------------
 
+// Quantize semantics (pinned, test-verified): q = clamp(rne(x * inv_scale
+// + zp)) with inv_scale = float(1.0/scale) and EXPLICITLY unfused mul/add
+// (__fmul_rn/__fadd_rn) — torch's scalar division also multiplies by the
+// reciprocal, and an fma contraction flips rare half-way ties.
 __global__ void quantize_q8_kernel(const float* __restrict__ in,
                                    int8_t* __restrict__ out, int64_t n,
-                                   float scale, float zero_point) {
+                                   float inv_scale, float zero_point) {
   int64_t i0 = (int64_t(blockIdx.x) * blockDim.x + threadIdx.x) * 4;
   int64_t stride = int64_t(gridDim.x) * blockDim.x * 4;
   for (int64_t i = i0; i < n; i += stride) {
     if (i + 4 <= n) {
       float4 v = *reinterpret_cast<const float4*>(in + i);
       char4 q;
-      q.x = int8_t(max(-128.f, min(127.f, nearbyintf(v.x / scale + zero_point))));
-      q.y = int8_t(max(-128.f, min(127.f, nearbyintf(v.y / scale + zero_point))));
-      q.z = int8_t(max(-128.f, min(127.f, nearbyintf(v.z / scale + zero_point))));
-      q.w = int8_t(max(-128.f, min(127.f, nearbyintf(v.w / scale + zero_point))));
+      q.x = int8_t(max(-128.f, min(127.f, nearbyintf(__fadd_rn(__fmul_rn(v.x, inv_scale), zero_point)))));
+      q.y = int8_t(max(-128.f, min(127.f, nearbyintf(__fadd_rn(__fmul_rn(v.y, inv_scale), zero_point)))));
+      q.z = int8_t(max(-128.f, min(127.f, nearbyintf(__fadd_rn(__fmul_rn(v.z, inv_scale), zero_point)))));
+      q.w = int8_t(max(-128.f, min(127.f, nearbyintf(__fadd_rn(__fmul_rn(v.w, inv_scale), zero_point)))));
       *reinterpret_cast<char4*>(out + i) = q;
     } else {
       for (int64_t k = i; k < n; ++k) {
-        float q = nearbyintf(in[k] / scale + zero_point);
+        float q = nearbyintf(__fadd_rn(__fmul_rn(in[k], inv_scale), zero_point));
         out[k] = int8_t(max(-128.f, min(127.f, q)));
       }
     }
@@ -605,7 +609,7 @@ at::Tensor quantize_q8(const at::Tensor& in, double scale,
   hipLaunchKernelGGL(quantize_q8_kernel, dim3(grid), dim3(block), 0,
                      current_stream(), in.data_ptr<float>(),
                      out.data_ptr<int8_t>(), n,
-                     float(scale), float(zero_point));
+                     float(1.0 / scale), float(zero_point));
   HIP_CHECK(hipGetLastError());
   return out;
 }

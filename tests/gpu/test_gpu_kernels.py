@@ -93,7 +93,11 @@ def test_quantize_q8_matches_torch(native):
     x = torch.randn(1 << 18, device=DEV) * 10
     scale, zp = 0.1, 3.0
     q = ops.quantize_q8(x, scale, zp)
-    ref = torch.clamp(torch.round(x / scale + zp), -128, 127).to(torch.int8)
+    # pinned semantics: multiply by float(1/scale), separate add, RNE —
+    # expressed identically on the torch side (torch's own x/scale also
+    # multiplies by the reciprocal but may differ at half-way ties)
+    inv = float(1.0 / scale)
+    ref = torch.clamp(torch.round(x * inv + zp), -128, 127).to(torch.int8)
     assert torch.equal(q, ref)
 
 
