@@ -25,7 +25,10 @@ ext = CUDAExtension(
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
-        "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
+        "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950",
+                 # numerics are pinned to torch's unfused mul/add; hipcc
+                 # contracts even __fmul_rn/__fadd_rn into fma otherwise
+                 "-ffp-contract=off"],
     },
 )
 
