@@ -100,6 +100,9 @@ def main():
                     choices=["echo", "resnet50", "bert"],
                     help="what the loopback server runs: echo (the codec/"
                          "transport benchmark) or a real model family")
+    ap.add_argument("--shards", type=int, default=1,
+                    help="split each logical request along dim 0 into this "
+                         "many parallel rpcs over separate channels")
     ap.add_argument("--pipeline", type=int, default=1,
                     help="in-flight requests per rank (1 = sequential; "
                          ">1 overlaps serialize/transport/parse of "
@@ -158,10 +161,15 @@ def main():
     if args.encoding == "turbo":
         from min_tfs_client_amd.turbo import TurboPredictClient
         client = TurboPredictClient(
-            address, num_channels=min(args.pipeline, 8))
+            address,
+            num_channels=max(min(args.pipeline, 8), args.shards))
 
         def step_fn(step_inputs):
             out_dev = f"cuda:{local_rank}" if has_gpu else "cpu"
+            if args.shards > 1:
+                return client.predict_sharded(
+                    "default", step_inputs, shards=args.shards,
+                    output_device=out_dev, copy_mode=args.copy_mode)
             return client.predict("default", step_inputs,
                                   output_device=out_dev,
                                   copy_mode=args.copy_mode)
@@ -312,6 +320,7 @@ def main():
                     next(iter(inputs.values())).shape),
                 "parallelism": f"dp{n_gpus}",
                 "pipeline": args.pipeline,
+                "shards": args.shards,
                 "encoding": args.encoding,
                 "copy_mode": args.copy_mode,
                 "transport": args.transport,

@@ -128,6 +128,57 @@ class TurboPredictClient:
                 resp, dev, copy_mode)
         return outputs
 
+    def predict_sharded(self, model_name: str,
+                        inputs: Dict[str, "torch.Tensor"],
+                        shards: int = 2,
+                        timeout: float = 60.0,
+                        model_version: Optional[int] = None,
+                        signature_name: str = "",
+                        output_device: Optional[str] = None,
+                        copy_mode: int = 1) -> Dict[str, "torch.Tensor"]:
+        """One logical Predict split along dim 0 into `shards` parallel
+        rpcs over separate channels: transport for a single request is
+        parallelized (each shard's serialize/send/recv/parse overlaps the
+        others), then outputs are re-concatenated along dim 0. Every input
+        must share its dim-0 size; uneven tails are handled. Falls back to
+        plain predict when sharding is not applicable. Wire-compatible:
+        the server just sees `shards` ordinary PredictRequests.
+        """
+        if shards <= 1 or len(self._stubs) < 2:
+            return self.predict(model_name, inputs, timeout, model_version,
+                                signature_name, output_device, copy_mode)
+        keys = list(inputs.keys())
+        batch = inputs[keys[0]].shape[0]
+        if any(inputs[k].dim() == 0 or inputs[k].shape[0] != batch
+               for k in keys) or batch < shards:
+            return self.predict(model_name, inputs, timeout, model_version,
+                                signature_name, output_device, copy_mode)
+        base, rem = divmod(batch, shards)
+        sizes = [base + (1 if i < rem else 0) for i in range(shards)]
+        futs = []
+        off = 0
+        for i, n in enumerate(sizes):
+            shard = {k: inputs[k].narrow(0, off, n) for k in keys}
+            blob = self.serialize_request(model_name, shard, model_version,
+                                          signature_name, copy_mode)
+            futs.append(self._stubs[i % len(self._stubs)].future(blob,
+                                                                 timeout))
+            off += n
+        dev = str(output_device) if output_device is not None else "cpu"
+        parts = []
+        for fut in futs:
+            _s, outputs, _ = self._native.parse_predict_response(
+                fut.result(), dev, copy_mode)
+            parts.append(outputs)
+        merged = {}
+        for k in parts[0]:
+            vals = [p[k] for p in parts]
+            if isinstance(vals[0], torch.Tensor) and vals[0].dim() > 0:
+                merged[k] = torch.cat(vals, dim=0)
+            else:
+                merged[k] = vals[0]
+        return merged
+
     def predict_future(self, model_name, inputs, timeout=60.0,
                        model_version=None, signature_name="",
                        copy_mode: int = 1):

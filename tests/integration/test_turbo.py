@@ -112,3 +112,24 @@ def test_turbo_multi_input_bert_shapes(raw_server):
                              {"input_ids": ids, "attention_mask": mask})
         assert torch.equal(out["input_ids"], ids)
         assert torch.equal(out["attention_mask"], mask)
+
+
+def test_predict_sharded(raw_server):
+    with TurboPredictClient(raw_server.address, num_channels=2) as client:
+        x = torch.randn(7, 3, 8, 8)  # uneven over 2 shards
+        y = torch.arange(7, dtype=torch.int64)
+        out = client.predict_sharded("default", {"x": x, "y": y}, shards=2)
+        assert torch.equal(out["x"], x)
+        assert torch.equal(out["y"], y)
+
+
+def test_predict_sharded_fallbacks(raw_server):
+    with TurboPredictClient(raw_server.address, num_channels=2) as client:
+        # scalar input: falls back to plain predict
+        out = client.predict_sharded("default", {"s": torch.tensor(2.5)},
+                                     shards=4)
+        assert out["s"].item() == 2.5
+        # batch smaller than shard count
+        out = client.predict_sharded("default", {"x": torch.ones(1, 3)},
+                                     shards=4)
+        assert torch.equal(out["x"], torch.ones(1, 3))
