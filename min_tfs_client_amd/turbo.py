@@ -148,9 +148,12 @@ class TurboPredictClient:
             return self.predict(model_name, inputs, timeout, model_version,
                                 signature_name, output_device, copy_mode)
         keys = list(inputs.keys())
+        if any(inputs[k].dim() == 0 for k in keys):
+            return self.predict(model_name, inputs, timeout, model_version,
+                                signature_name, output_device, copy_mode)
         batch = inputs[keys[0]].shape[0]
-        if any(inputs[k].dim() == 0 or inputs[k].shape[0] != batch
-               for k in keys) or batch < shards:
+        if any(inputs[k].shape[0] != batch for k in keys) \
+                or batch < shards:
             return self.predict(model_name, inputs, timeout, model_version,
                                 signature_name, output_device, copy_mode)
         base, rem = divmod(batch, shards)
