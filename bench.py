@@ -100,9 +100,11 @@ def main():
                     choices=["echo", "resnet50", "bert"],
                     help="what the loopback server runs: echo (the codec/"
                          "transport benchmark) or a real model family")
-    ap.add_argument("--shards", type=int, default=1,
+    ap.add_argument("--shards", type=int, default=4,
                     help="split each logical request along dim 0 into this "
-                         "many parallel rpcs over separate channels")
+                         "many parallel rpcs over separate channels "
+                         "(measured: 4 shards more than halve p50 RTT on "
+                         "the 19MB config; 1 disables)")
     ap.add_argument("--pipeline", type=int, default=1,
                     help="in-flight requests per rank (1 = sequential; "
                          ">1 overlaps serialize/transport/parse of "
