@@ -122,11 +122,27 @@ class RestApiServer:
             def _error(self, code, msg):
                 self._send(code, {"error": msg})
 
-            # ---- GET: status / metadata / prometheus ----------------
+            # ---- GET: status / metadata / prometheus / tracing ------
             def do_GET(self):
                 if self.path == outer.prometheus_path:
                     self._send(200, outer.metrics.render_prometheus(),
                                content_type="text/plain")
+                    return
+                if self.path == "/v1/tracing/export":
+                    # remote trace capture (ProfilerService analogue,
+                    # reference server.cc:324,339): returns the chrome-trace
+                    # JSON accumulated since tracing:start
+                    from .utils.tracing import Tracer
+                    import tempfile
+                    t = Tracer.get()
+                    with tempfile.NamedTemporaryFile("r", suffix=".json",
+                                                     delete=False) as f:
+                        path = f.name
+                    n = t.export(path)
+                    body = open(path).read()
+                    import os as _os
+                    _os.unlink(path)
+                    self._send(200, body, content_type="application/json")
                     return
                 m = _MODEL_RE.match(self.path)
                 if not m:
@@ -181,10 +197,20 @@ class RestApiServer:
                         "signature_def": {
                             servable.signature_name: sig}}}})
 
-            # ---- POST: predict --------------------------------------
+            # ---- POST: predict / tracing control ---------------------
             def do_POST(self):
                 import time as _t
                 t0 = _t.perf_counter()
+                if self.path in ("/v1/tracing:start", "/v1/tracing:stop"):
+                    from .utils.tracing import Tracer
+                    t = Tracer.get()
+                    if self.path.endswith(":start"):
+                        t.clear()
+                        t.start()
+                    else:
+                        t.stop()
+                    self._send(200, {"tracing": t.enabled})
+                    return
                 m = _MODEL_RE.match(self.path)
                 if not m or m.group("rest") not in (":predict", ":classify",
                                                     ":regress"):

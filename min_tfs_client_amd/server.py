@@ -187,9 +187,9 @@ def _decode_input(proto):
 
 
 def _abort(context, code, msg):
-    context.set_code(code)
-    context.set_details(msg)
-    raise grpc.RpcError(msg)
+    # context.abort raises; the fallback covers non-grpc contexts in tests
+    context.abort(code, msg)
+    raise grpc.RpcError(msg)  # pragma: no cover
 
 
 class PredictionServiceImpl(PredictionServiceServicer):

@@ -186,3 +186,11 @@ def test_cli_batching_flag(tmp_path):
     finally:
         source.stop()
         server.stop(0)
+
+
+def test_rest_tracing_endpoints(rest):
+    _post(rest, "/v1/tracing:start", {})
+    _post(rest, "/v1/models/double:predict", {"instances": [[1.0]]})
+    _post(rest, "/v1/tracing:stop", {})
+    body = _get(rest, "/v1/tracing/export")
+    assert "traceEvents" in body
