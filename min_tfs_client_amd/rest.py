@@ -218,11 +218,11 @@ class RestApiServer:
                     return
                 name = m.group("model")
                 version = m.group("version")
+                method = m.group("rest")
                 length = int(self.headers.get("Content-Length", 0))
                 try:
                     payload = json.loads(self.rfile.read(length) or b"{}")
-                    inputs, row_format = _json_to_inputs(payload)
-                except (ValueError, json.JSONDecodeError) as e:
+                except json.JSONDecodeError as e:
                     self._error(400, str(e))
                     return
                 try:
@@ -230,6 +230,17 @@ class RestApiServer:
                         name, int(version) if version else None)
                 except KeyError as e:
                     self._error(404, str(e))
+                    return
+                if method in (":classify", ":regress"):
+                    self._classify_regress(servable, method, payload)
+                    outer.metrics.observe_request(
+                        "rest" + method.replace(":", "_"),
+                        _t.perf_counter() - t0)
+                    return
+                try:
+                    inputs, row_format = _json_to_inputs(payload)
+                except ValueError as e:
+                    self._error(400, str(e))
                     return
                 try:
                     outputs = servable(inputs)
@@ -239,6 +250,41 @@ class RestApiServer:
                 self._send(200, _outputs_to_json(outputs, row_format))
                 outer.metrics.observe_request("rest_predict",
                                               _t.perf_counter() - t0)
+
+            def _classify_regress(self, servable, method, payload):
+                """TF REST classify/regress format
+                (http_rest_api_handler.h:63-81): request {"examples":
+                [{feat: value}, ...]}; classify responds {"results":
+                [[[label, score], ...], ...]}, regress {"results":
+                [v, ...]}."""
+                from .examples_adapter import examples_input
+                examples = payload.get("examples")
+                if not isinstance(examples, list) or not examples:
+                    self._error(400, 'request must carry "examples"')
+                    return
+                try:
+                    input_proto = examples_input(examples)
+                except (TypeError, ValueError) as e:
+                    self._error(400, str(e))
+                    return
+                fn_name = "classify" if method == ":classify" else "regress"
+                fn = getattr(servable, fn_name, None)
+                if fn is None:
+                    self._error(
+                        400, f"Expected a {fn_name[:-1] if 0 else fn_name}"
+                             f" signature for this model")
+                    return
+                try:
+                    result = fn(input_proto)
+                except Exception as e:  # noqa: BLE001
+                    self._error(500, str(e))
+                    return
+                if method == ":classify":
+                    results = [[[c.label, c.score] for c in cls.classes]
+                               for cls in result.classifications]
+                else:
+                    results = [r.value for r in result.regressions]
+                self._send(200, {"results": results})
 
         self._httpd = ThreadingHTTPServer(("127.0.0.1", port), Handler)
         self.port = self._httpd.server_port

@@ -194,3 +194,28 @@ def test_rest_tracing_endpoints(rest):
     _post(rest, "/v1/tracing:stop", {})
     body = _get(rest, "/v1/tracing/export")
     assert "traceEvents" in body
+
+
+def test_rest_classify_and_regress(manager):
+    from min_tfs_client_amd.examples_adapter import (
+        ClassificationAdapter, RegressionAdapter)
+    from min_tfs_client_amd.server import Servable
+
+    def fn(features):
+        s = features["x"].sum(axis=1)
+        return {"scores": np.stack([s, -s], 1), "value": s}
+
+    manager.load("clf", ClassificationAdapter(Servable(fn),
+                                              labels=["a", "b"]), version=1)
+    manager.load("reg", RegressionAdapter(Servable(fn)), version=1)
+    with RestApiServer(manager, port=0) as rest_srv:
+        body = _post(rest_srv, "/v1/models/clf:classify",
+                     {"examples": [{"x": [1.0, 2.0]}]})
+        assert body == {"results": [[["a", 3.0], ["b", -3.0]]]}
+        body = _post(rest_srv, "/v1/models/reg:regress",
+                     {"examples": [{"x": [2.0, 2.0]}, {"x": [1.0, 0.0]}]})
+        assert body["results"][0] == 4.0
+        with pytest.raises(urllib.error.HTTPError) as err:
+            _post(rest_srv, "/v1/models/default:classify",
+                  {"examples": [{"x": [1.0]}]})
+        assert err.value.code == 400
