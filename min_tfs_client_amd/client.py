@@ -78,10 +78,13 @@ class TensorServingClient:
     # ------------------------------------------------------------------
     def _fill_model_spec(self, spec, model_name: str,
                          model_version: Optional[int],
-                         signature_name: Optional[str] = None) -> None:
+                         signature_name: Optional[str] = None,
+                         version_label: Optional[str] = None) -> None:
         spec.name = model_name
         if model_version is not None:
             spec.version.value = model_version
+        elif version_label:
+            spec.version_label = version_label
         if signature_name:
             spec.signature_name = signature_name
 
@@ -102,10 +105,11 @@ class TensorServingClient:
         signature_name: Optional[str] = None,
         output_filter: Optional[list] = None,
         use_tensor_content: bool = True,
+        version_label: Optional[str] = None,
     ) -> "pb.PredictResponse":
         request = pb.PredictRequest()
         self._fill_model_spec(request.model_spec, model_name, model_version,
-                              signature_name)
+                              signature_name, version_label)
         for k, v in input_dict.items():
             request.inputs[k].CopyFrom(
                 self._encode_input(v, use_tensor_content))

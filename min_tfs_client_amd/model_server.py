@@ -113,7 +113,7 @@ def make_server(args) -> tuple:
         fail_if_zero_versions_at_startup=(
             args.fail_if_zero_versions_at_startup))
 
-    configs, policies = {}, {}
+    configs, policies, labels = {}, {}, {}
     if args.model_config_file:
         cfg = _parse_text_proto(args.model_config_file, pb.ModelServerConfig)
         for mc in cfg.model_config_list.config:
@@ -121,9 +121,16 @@ def make_server(args) -> tuple:
             policies[mc.name] = VersionPolicy.from_proto(
                 mc.model_version_policy
                 if mc.HasField("model_version_policy") else None)
+            if mc.version_labels:
+                labels[mc.name] = dict(mc.version_labels)
     elif args.model_base_path:
         configs[args.model_name] = args.model_base_path
     source.set_models(configs, policies)
+    if labels:
+        source.poll_once()  # versions must be AVAILABLE before labeling
+        for name, lbls in labels.items():
+            for label, ver in lbls.items():
+                server.manager.set_version_label(name, label, ver)
 
     rest = None
     if args.rest_api_port:

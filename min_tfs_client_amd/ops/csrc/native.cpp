@@ -214,6 +214,7 @@ py::tuple parse_predict(py::buffer data, bool is_request,
   spec["name"] = parsed.model_spec.name;
   spec["version"] = parsed.model_spec.version;
   spec["signature_name"] = parsed.model_spec.signature_name;
+  spec["version_label"] = parsed.model_spec.version_label;
   py::list filt;
   for (auto& f : parsed.output_filter) filt.append(py::str(f));
   return py::make_tuple(spec, out, filt);

@@ -297,6 +297,7 @@ struct ParsedModelSpec {
   std::string name;
   int64_t version = -1;
   std::string signature_name;
+  std::string version_label;   // ModelSpec field 4 (oneof with version)
 };
 
 inline void parse_shape(Cursor c, std::vector<int64_t>* shape) {
@@ -395,6 +396,10 @@ inline void parse_model_spec(Cursor c, ParsedModelSpec* m) {
       Cursor b = c.read_len_delim();
       m->signature_name.assign(reinterpret_cast<const char*>(b.p),
                                size_t(b.end - b.p));
+    } else if (f == 4 && wt == WT_LEN) {
+      Cursor b = c.read_len_delim();
+      m->version_label.assign(reinterpret_cast<const char*>(b.p),
+                              size_t(b.end - b.p));
     } else {
       c.skip(wt);
     }

@@ -109,6 +109,9 @@ class ModelManager:
     def __init__(self):
         self._lock = threading.RLock()
         self._models: Dict[str, Dict[int, _Version]] = {}
+        # {model: {label: version}} — ModelSpec.version_label routing
+        # (model.proto:9-33 oneof version_choice; ModelConfig.version_labels)
+        self._labels: Dict[str, Dict[str, int]] = {}
 
     # -- lifecycle ------------------------------------------------------
     def load(self, name: str, servable: Servable, version: int = 1) -> None:
@@ -139,13 +142,33 @@ class ModelManager:
                     versions[ver].state = STATE_END
                     versions[ver].servable = None
 
+    def set_version_label(self, name: str, label: str,
+                          version: int) -> None:
+        """Labels may only point at AVAILABLE versions (TF semantics:
+        --allow_version_labels_for_unavailable_models defaults false)."""
+        with self._lock:
+            v = self._models.get(name, {}).get(version)
+            if v is None or v.state != STATE_AVAILABLE:
+                raise KeyError(
+                    f"Request to assign label to version {version} of model "
+                    f"{name}, which is not currently available for "
+                    f"inference")
+            self._labels.setdefault(name, {})[label] = version
+
     # -- lookup ---------------------------------------------------------
-    def get(self, name: str, version: Optional[int] = None) -> Servable:
+    def get(self, name: str, version: Optional[int] = None,
+            version_label: Optional[str] = None) -> Servable:
         """Resolve to an AVAILABLE servable; raises KeyError with a
         TF-Serving-shaped message."""
         with self._lock:
             if name not in self._models:
                 raise KeyError(f"Servable not found for request: Latest({name})")
+            if version is None and version_label:
+                version = self._labels.get(name, {}).get(version_label)
+                if version is None:
+                    raise KeyError(
+                        f"Servable not found for request: Specific({name}, "
+                        f"label {version_label})")
             versions = self._models[name]
             if version is not None:
                 v = versions.get(version)
@@ -206,10 +229,14 @@ class PredictionServiceImpl(PredictionServiceServicer):
     # -- helpers --------------------------------------------------------
     def _resolve(self, model_spec, context):
         version = None
-        if model_spec.HasField("version"):
+        label = None
+        which = model_spec.WhichOneof("version_choice")
+        if which == "version":
             version = model_spec.version.value
+        elif which == "version_label":
+            label = model_spec.version_label
         try:
-            return self._manager.get(model_spec.name, version)
+            return self._manager.get(model_spec.name, version, label)
         except KeyError as e:
             _abort(context, grpc.StatusCode.NOT_FOUND, str(e))
 
@@ -413,8 +440,9 @@ def _raw_predict_handler(manager: ModelManager, device: str,
             _abort(context, grpc.StatusCode.INVALID_ARGUMENT,
                    f"request parsing error: {e}")
         version = spec["version"] if spec["version"] >= 0 else None
+        label = spec.get("version_label") or None
         try:
-            servable = manager.get(spec["name"], version)
+            servable = manager.get(spec["name"], version, label)
         except KeyError as e:
             _abort(context, grpc.StatusCode.NOT_FOUND, str(e))
         if getattr(servable, "is_identity", False):

@@ -178,3 +178,25 @@ def test_enable_retries_option(server):
             tensor_proto_to_ndarray(resp.outputs["x"]), x)
     finally:
         c.close()
+
+
+def test_version_label_routing(server):
+    """ModelSpec.version_label resolution (model.proto oneof
+    version_choice; ModelConfig.version_labels semantics)."""
+    server.manager.load("labeled", identity_servable(), version=1)
+    server.manager.load("labeled", identity_servable(), version=2)
+    server.manager.set_version_label("labeled", "stable", 1)
+    server.manager.set_version_label("labeled", "canary", 2)
+    c = TensorServingClient("127.0.0.1", server.port)
+    try:
+        x = np.ones(2, np.float32)
+        r = c.predict_request("labeled", {"x": x}, version_label="stable")
+        assert "x" in r.outputs
+        with pytest.raises(grpc.RpcError) as err:
+            c.predict_request("labeled", {"x": x}, version_label="nope")
+        assert err.value.code() == grpc.StatusCode.NOT_FOUND
+        # labels may only target AVAILABLE versions
+        with pytest.raises(KeyError):
+            server.manager.set_version_label("labeled", "bad", 99)
+    finally:
+        c.close()

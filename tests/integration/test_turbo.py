@@ -133,3 +133,23 @@ def test_predict_sharded_fallbacks(raw_server):
         out = client.predict_sharded("default", {"x": torch.ones(1, 3)},
                                      shards=4)
         assert torch.equal(out["x"], torch.ones(1, 3))
+
+
+def test_turbo_version_label_via_raw_server(raw_server):
+    """version_label travels through the C++ parse path too (ModelSpec
+    field 4)."""
+    raw_server.manager.load("lbl", identity_servable(), version=3)
+    raw_server.manager.set_version_label("lbl", "prod", 3)
+    from min_tfs_client_amd.wire import messages as pb
+    from min_tfs_client_amd import _native as native
+    req = pb.PredictRequest()
+    req.model_spec.name = "lbl"
+    req.model_spec.version_label = "prod"
+    import numpy as np
+    from min_tfs_client_amd.tensors import ndarray_to_tensor_proto
+    req.inputs["x"].CopyFrom(ndarray_to_tensor_proto(
+        np.ones(2, np.float32)))
+    with TurboPredictClient(raw_server.address) as c:
+        resp = c._predict(req.SerializeToString(), 30)
+        _s, outs, _ = native.parse_predict_response(resp, "cpu", 1)
+        assert torch.equal(outs["x"], torch.ones(2))
