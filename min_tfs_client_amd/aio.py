@@ -1,0 +1,73 @@
+"""Asyncio Predict client over the C++ codec (grpc.aio transport).
+
+The sync ``TurboPredictClient`` pipelines with futures; this variant gives
+the same hot path to asyncio servers/apps: ``await client.predict(...)``,
+natural fan-out with ``asyncio.gather``. Serialize/parse run in the C++
+extension with the GIL released, so event-loop stalls stay sub-ms even at
+19 MB payloads.
+"""
+from __future__ import annotations
+
+import asyncio
+from typing import Dict, Optional
+
+import grpc
+import grpc.aio
+
+try:
+    import torch
+except ImportError:  # pragma: no cover
+    torch = None
+
+from .ops import require_native
+from .turbo import _CHANNEL_OPTS, _PREDICT_PATH, _identity
+from .utils.allocator import tune_malloc
+
+
+class AsyncTurboPredictClient:
+    def __init__(self, target: str,
+                 credentials: Optional[grpc.ChannelCredentials] = None,
+                 options: Optional[list] = None):
+        self._native = require_native()
+        tune_malloc()
+        opts = _CHANNEL_OPTS + (options or [])
+        if credentials:
+            self._channel = grpc.aio.secure_channel(target, credentials,
+                                                    options=opts)
+        else:
+            self._channel = grpc.aio.insecure_channel(target, options=opts)
+        self._predict = self._channel.unary_unary(
+            _PREDICT_PATH, request_serializer=_identity,
+            response_deserializer=_identity)
+
+    async def close(self):
+        await self._channel.close()
+
+    async def __aenter__(self):
+        return self
+
+    async def __aexit__(self, *exc):
+        await self.close()
+
+    async def predict(self, model_name: str,
+                      inputs: Dict[str, "torch.Tensor"],
+                      timeout: float = 60.0,
+                      model_version: Optional[int] = None,
+                      signature_name: str = "",
+                      output_device: Optional[str] = None,
+                      copy_mode: int = 1) -> Dict[str, "torch.Tensor"]:
+        names = list(inputs.keys())
+        tensors = [inputs[k] for k in names]
+        loop = asyncio.get_running_loop()
+        # serialize in the default executor: the C++ side releases the GIL
+        # during copies but the call itself can take ~1 ms at 19 MB
+        blob = await loop.run_in_executor(
+            None, lambda: self._native.serialize_predict_request(
+                model_name, -1 if model_version is None else model_version,
+                signature_name, names, tensors, copy_mode))
+        resp = await self._predict(blob, timeout=timeout)
+        dev = str(output_device) if output_device is not None else "cpu"
+        _spec, outputs, _ = await loop.run_in_executor(
+            None, lambda: self._native.parse_predict_response(
+                resp, dev, copy_mode))
+        return outputs

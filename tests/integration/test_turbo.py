@@ -153,3 +153,18 @@ def test_turbo_version_label_via_raw_server(raw_server):
         resp = c._predict(req.SerializeToString(), 30)
         _s, outs, _ = native.parse_predict_response(resp, "cpu", 1)
         assert torch.equal(outs["x"], torch.ones(2))
+
+
+def test_async_client(raw_server):
+    import asyncio
+    from min_tfs_client_amd.aio import AsyncTurboPredictClient
+
+    async def run():
+        async with AsyncTurboPredictClient(raw_server.address) as c:
+            xs = [torch.randn(4, 4) for _ in range(6)]
+            outs = await asyncio.gather(
+                *[c.predict("default", {"x": x}) for x in xs])
+            for out, x in zip(outs, xs):
+                assert torch.equal(out["x"], x)
+
+    asyncio.run(run())
