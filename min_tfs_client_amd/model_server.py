@@ -77,12 +77,13 @@ def _parse_text_proto(path: str, cls):
 
 def make_server(args) -> tuple:
     """Builds (ModelServer, FileSystemStoragePathSource, RestApiServer?)."""
-    loader = default_loader
+    import functools
+    loader = functools.partial(default_loader, device=args.device)
     if args.enable_batching:
         allowed = ([int(x) for x in args.allowed_batch_sizes.split(",")]
                    if args.allowed_batch_sizes else None)
 
-        def loader(name, vdir, _inner=default_loader):  # noqa: F811
+        def loader(name, vdir, _inner=loader):  # noqa: F811
             return BatchingServable(
                 _inner(name, vdir),
                 max_batch_size=args.max_batch_size,
@@ -92,7 +93,7 @@ def make_server(args) -> tuple:
     server = ModelServer(port=args.port, raw_predict=args.raw_predict,
                          device=args.device,
                          servable_factory=lambda name, path:
-                         default_loader(name, path))
+                         default_loader(name, path, device=args.device))
     if args.grpc_socket_path:
         server._server.add_insecure_port(f"unix://{args.grpc_socket_path}")
     if args.ssl_config_file:

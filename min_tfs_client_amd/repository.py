@@ -142,9 +142,11 @@ def run_warmup(servable: Servable, version_dir: str,
 # Loaders
 # ---------------------------------------------------------------------------
 
-def default_loader(name: str, version_dir: str) -> Servable:
+def default_loader(name: str, version_dir: str,
+                   device: str = "cpu") -> Servable:
     """Builds a Servable from a version directory (formats in module
-    docstring)."""
+    docstring). ``device`` is the default placement for formats that do
+    not name one themselves (model.json may override it)."""
     if os.path.exists(os.path.join(version_dir, "identity")) or \
             os.path.exists(os.path.join(version_dir, "saved_model.pb")):
         # saved_model.pb acceptance keeps reference-style fixture layouts
@@ -155,7 +157,7 @@ def default_loader(name: str, version_dir: str) -> Servable:
         with open(cfg_path) as f:
             cfg = json.load(f)
         family = cfg.get("family")
-        device = cfg.get("device", "cpu")
+        device = cfg.get("device", device)
         if family == "resnet50":
             from .models import resnet50_servable
             servable = resnet50_servable(device)
@@ -181,14 +183,15 @@ def default_loader(name: str, version_dir: str) -> Servable:
     pt_path = os.path.join(version_dir, "model.pt")
     if os.path.exists(pt_path):
         import torch
-        module = torch.jit.load(pt_path, map_location="cpu")
+        module = torch.jit.load(pt_path, map_location=device)
         module.eval()
 
         @torch.no_grad()
         def fn(inputs):
             keys = sorted(inputs.keys())
-            tensors = [inputs[k] if isinstance(inputs[k], torch.Tensor)
-                       else torch.as_tensor(inputs[k]) for k in keys]
+            tensors = [(inputs[k] if isinstance(inputs[k], torch.Tensor)
+                        else torch.as_tensor(inputs[k])).to(device)
+                       for k in keys]
             out = module(*tensors)
             if isinstance(out, dict):
                 return out
