@@ -25,3 +25,20 @@ from .tensors import (  # noqa: F401
     extract_shape,
 )
 from .types import DataType  # noqa: F401
+
+
+def __getattr__(name):
+    # heavier entry points, imported lazily to keep bare import cheap
+    if name == "TurboPredictClient":
+        from .turbo import TurboPredictClient
+        return TurboPredictClient
+    if name == "AsyncTurboPredictClient":
+        from .aio import AsyncTurboPredictClient
+        return AsyncTurboPredictClient
+    if name == "ModelServer":
+        from .server import ModelServer
+        return ModelServer
+    if name == "DataParallelPredictor":
+        from .parallel import DataParallelPredictor
+        return DataParallelPredictor
+    raise AttributeError(name)

@@ -270,9 +270,8 @@ class RestApiServer:
                 fn_name = "classify" if method == ":classify" else "regress"
                 fn = getattr(servable, fn_name, None)
                 if fn is None:
-                    self._error(
-                        400, f"Expected a {fn_name[:-1] if 0 else fn_name}"
-                             f" signature for this model")
+                    self._error(400, f"Expected a {fn_name} signature "
+                                     f"for this model")
                     return
                 try:
                     result = fn(input_proto)
