@@ -114,32 +114,33 @@ struct F16ToBf16 {
 // elementwise cast, 8 elements per lane, 16B loads where possible (G13)
 // ---------------------------------------------------------------------------
 
-template <typename CVT>
+template <typename CVT, int UNROLL = 2>
 __global__ void cast_kernel(const typename CVT::In* __restrict__ in,
                             typename CVT::Out* __restrict__ out,
                             int64_t n) {
   using In = typename CVT::In;
   using Out = typename CVT::Out;
-  constexpr int V = 16 / sizeof(In);  // elements per 16B load
+  constexpr int VC = 16 / sizeof(In);      // elements per 16B chunk
+  constexpr int V = VC * UNROLL;           // elements per lane per iter
   int64_t i0 = (int64_t(blockIdx.x) * blockDim.x + threadIdx.x) * V;
   int64_t stride = int64_t(gridDim.x) * blockDim.x * V;
   for (int64_t i = i0; i < n; i += stride) {
     if (i + V <= n) {
       In vin[V];
-      *reinterpret_cast<int4*>(vin) =
-          *reinterpret_cast<const int4*>(in + i);
+#pragma unroll
+      for (int u = 0; u < UNROLL; ++u)
+        reinterpret_cast<int4*>(vin)[u] =
+            reinterpret_cast<const int4*>(in + i)[u];
       Out vout[V];
 #pragma unroll
       for (int k = 0; k < V; ++k) vout[k] = CVT::cvt(vin[k]);
-      if constexpr (sizeof(Out) * V == 16) {
-        *reinterpret_cast<int4*>(out + i) =
-            *reinterpret_cast<const int4*>(vout);
-      } else if constexpr (sizeof(Out) * V == 32) {
-        *reinterpret_cast<int4*>(out + i) =
-            *reinterpret_cast<const int4*>(vout);
-        *reinterpret_cast<int4*>(out + i + V / 2) =
-            *reinterpret_cast<const int4*>(vout + V / 2);
-      } else {  // 8 bytes out
+      constexpr int OUTB = int(sizeof(Out)) * V;
+      if constexpr (OUTB % 16 == 0) {
+#pragma unroll
+        for (int u = 0; u < OUTB / 16; ++u)
+          reinterpret_cast<int4*>(out + i)[u] =
+              reinterpret_cast<const int4*>(vout)[u];
+      } else {  // 8-byte tail width (16B in -> 8B out, UNROLL=1)
         *reinterpret_cast<int2*>(out + i) =
             *reinterpret_cast<const int2*>(vout);
       }
@@ -408,11 +409,12 @@ static hipStream_t current_stream() {
 template <typename CVT>
 static void launch_cast(const at::Tensor& in, at::Tensor& out) {
   const int64_t n = in.numel();
-  constexpr int V = 16 / sizeof(typename CVT::In);
+  constexpr int UNROLL = 2;  // 32B loads/lane: A/B'd vs 16B on MI355X
+  constexpr int V = UNROLL * 16 / sizeof(typename CVT::In);
   const int block = 256;
   const int grid = grid_for((n + V - 1) / V, block);
-  hipLaunchKernelGGL(cast_kernel<CVT>, dim3(grid), dim3(block), 0,
-                     current_stream(),
+  hipLaunchKernelGGL((cast_kernel<CVT, UNROLL>), dim3(grid), dim3(block),
+                     0, current_stream(),
                      reinterpret_cast<const typename CVT::In*>(
                          in.const_data_ptr()),
                      reinterpret_cast<typename CVT::Out*>(
