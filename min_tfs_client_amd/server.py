@@ -447,6 +447,8 @@ def _raw_predict_handler(manager: ModelManager, device: str,
             _abort(context, grpc.StatusCode.NOT_FOUND, str(e))
         if getattr(servable, "is_identity", False):
             out = native.echo_predict(data)
+            metrics.observe_bytes("rx", len(data))
+            metrics.observe_bytes("tx", len(out))
             metrics.observe_request("predict", time.perf_counter() - t0)
             if request_logger is not None:
                 request_logger.log_predict(spec["name"], bytes(data),

@@ -120,8 +120,10 @@ class TurboPredictClient:
                                   for t in inputs.values())):
             blob = self.serialize_request(model_name, inputs, model_version,
                                           signature_name, copy_mode)
+        self.metrics.observe_bytes("tx", len(blob))
         with trace_span("turbo.rpc", bytes=len(blob)):
             resp = self._predict(blob, timeout)
+        self.metrics.observe_bytes("rx", len(resp))
         dev = str(output_device) if output_device is not None else "cpu"
         with trace_span("turbo.parse", device=dev):
             _spec, outputs, _ = self._native.parse_predict_response(

@@ -168,3 +168,30 @@ def test_async_client(raw_server):
                 assert torch.equal(out["x"], x)
 
     asyncio.run(run())
+
+
+def test_deadline_exceeded():
+    """gRPC deadline semantics: timeout is seconds, slow servables abort
+    with DEADLINE_EXCEEDED (reference requests.py:49 passes timeout
+    positionally the same way)."""
+    import grpc
+    import time as _time
+
+    def slow(inputs):
+        _time.sleep(2.0)
+        return inputs
+
+    with ModelServer(port=0, raw_predict=True) as srv:
+        srv.manager.load("slow", Servable(slow), version=1)
+        with TurboPredictClient(srv.address) as c:
+            with pytest.raises(grpc.RpcError) as err:
+                c.predict("slow", {"x": torch.ones(1)}, timeout=0.3)
+            assert err.value.code() == grpc.StatusCode.DEADLINE_EXCEEDED
+
+
+def test_client_bytes_metrics(raw_server):
+    with TurboPredictClient(raw_server.address) as c:
+        c.predict("default", {"x": torch.ones(100)})
+        counters = c.metrics.counters()
+        assert counters["bytes_total{direction='tx'}"] > 400
+        assert counters["bytes_total{direction='rx'}"] > 400

@@ -106,3 +106,14 @@ def test_truncations_never_crash(data):
         native.parse_predict_request(bytes(mutated), "cpu", 0)
     except Exception:
         pass
+
+
+def test_uint_barebones_dtypes_roundtrip():
+    """torch's barebones uint16/32/64 dtypes map to DT_UINT16/32/64."""
+    for dt, enum in ((torch.uint16, 17), (torch.uint32, 22),
+                     (torch.uint64, 23)):
+        t = torch.arange(6).to(dt)
+        blob = native.serialize_predict_request("m", -1, "", ["x"], [t], 0)
+        assert pb.PredictRequest.FromString(blob).inputs["x"].dtype == enum
+        _s, outs, _ = native.parse_predict_request(blob, "cpu", 0)
+        assert outs["x"].dtype == dt and torch.equal(outs["x"], t)
