@@ -259,6 +259,9 @@ class FileSystemStoragePathSource:
         self.fail_if_zero = fail_if_zero_versions_at_startup
         self._configs: Dict[str, tuple] = {}  # name -> (base_path, policy)
         self._loaded: Dict[str, Dict[int, str]] = {}
+        # versions that exhausted their retries: not re-attempted until
+        # their directory disappears (TF gives up after max retries too)
+        self._failed: Dict[str, set] = {}
         self._lock = threading.Lock()
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
@@ -290,10 +293,14 @@ class FileSystemStoragePathSource:
                     f"model {name}: no versions at startup under {base}")
             aspired = set(policy.aspired(sorted(found)))
             loaded = self._loaded.setdefault(name, {})
-            for ver in sorted(aspired - set(loaded)):
+            failed = self._failed.setdefault(name, set())
+            failed.intersection_update(found)   # dir removed -> forget
+            for ver in sorted(aspired - set(loaded) - failed):
                 vdir = os.path.join(base, found[ver])
                 if self._load_with_retries(name, ver, vdir):
                     loaded[ver] = vdir
+                else:
+                    failed.add(ver)
             for ver in sorted(set(loaded) - aspired):
                 self.manager.unload(name, ver)
                 del loaded[ver]

@@ -445,7 +445,7 @@ def _raw_predict_handler(manager: ModelManager, device: str,
             servable = manager.get(spec["name"], version, label)
         except KeyError as e:
             _abort(context, grpc.StatusCode.NOT_FOUND, str(e))
-        if getattr(servable, "is_identity", False):
+        if getattr(servable, "is_identity", False) and not _filter:
             out = native.echo_predict(data)
             metrics.observe_bytes("rx", len(data))
             metrics.observe_bytes("tx", len(out))
@@ -458,6 +458,8 @@ def _raw_predict_handler(manager: ModelManager, device: str,
             outputs = servable(inputs)
         except Exception as e:  # noqa: BLE001
             _abort(context, grpc.StatusCode.INTERNAL, str(e))
+        if _filter:
+            outputs = {k: v for k, v in outputs.items() if k in _filter}
         names = list(outputs.keys())
         tensors = []
         for k in names:

@@ -195,3 +195,21 @@ def test_client_bytes_metrics(raw_server):
         counters = c.metrics.counters()
         assert counters["bytes_total{direction='tx'}"] > 400
         assert counters["bytes_total{direction='rx'}"] > 400
+
+
+def test_output_filter_raw_path(raw_server):
+    """output_filter must be honored on the raw C++ path too (for the
+    identity servable this disables the echo fast path)."""
+    from min_tfs_client_amd.wire import messages as pb
+    from min_tfs_client_amd import _native as native
+    import numpy as np
+    from min_tfs_client_amd.tensors import ndarray_to_tensor_proto
+    req = pb.PredictRequest()
+    req.model_spec.name = "default"
+    req.inputs["a"].CopyFrom(ndarray_to_tensor_proto(np.ones(1, np.float32)))
+    req.inputs["b"].CopyFrom(ndarray_to_tensor_proto(np.ones(1, np.float32)))
+    req.output_filter.append("a")
+    with TurboPredictClient(raw_server.address) as c:
+        resp_bytes = c._predict(req.SerializeToString(), 30)
+    resp = pb.PredictResponse.FromString(resp_bytes)
+    assert sorted(resp.outputs) == ["a"]

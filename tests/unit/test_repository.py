@@ -249,3 +249,31 @@ def test_loader_state_dict(tmp_path):
                                   donor.module.state_dict().items()):
         assert n1 == n2
         assert torch.equal(p1, p2)
+
+
+def test_failed_version_not_retried_until_dir_removed(tmp_path):
+    mgr = ModelManager()
+    attempts = []
+
+    def failing_loader(name, vdir):
+        attempts.append(1)
+        raise RuntimeError("permanent")
+
+    src = FileSystemStoragePathSource(mgr, loader=failing_loader,
+                                      poll_wait_seconds=0,
+                                      max_num_load_retries=1,
+                                      load_retry_interval_s=0)
+    _make_version(tmp_path, "m", 1)
+    src.set_models({"m": str(tmp_path / "m")})
+    src.poll_once()
+    n_after_first = len(attempts)
+    src.poll_once()
+    src.poll_once()
+    assert len(attempts) == n_after_first  # no re-attempts
+    # removing and recreating the version dir clears the failure memory
+    import shutil
+    shutil.rmtree(str(tmp_path / "m" / "1"))
+    src.poll_once()
+    _make_version(tmp_path, "m", 1)
+    src.poll_once()
+    assert len(attempts) > n_after_first
