@@ -236,6 +236,9 @@ def main():
 
     if args.pipeline > 1 and world_size > 1:
         raise SystemExit("--pipeline > 1 is single-rank only")
+    if args.pipeline > 1 and args.shards > 1:
+        raise SystemExit("--pipeline and --shards are mutually exclusive; "
+                         "pass --shards 1 with --pipeline")
     if args.pipeline > 1 and args.encoding != "turbo":
         raise SystemExit("--pipeline > 1 requires --encoding turbo")
 

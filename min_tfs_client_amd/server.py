@@ -139,6 +139,10 @@ class ModelManager:
             for ver in targets:
                 if ver in versions:
                     versions[ver].state = STATE_UNLOADING
+                    servable = versions[ver].servable
+                    close = getattr(servable, "close", None)
+                    if callable(close):
+                        close()  # e.g. BatchingServable's batcher thread
                     versions[ver].state = STATE_END
                     versions[ver].servable = None
 

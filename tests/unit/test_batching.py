@@ -138,3 +138,16 @@ def test_through_server():
             np.testing.assert_array_equal(
                 results[i], np.full((2, 2), i * 2, dtype=np.float32))
         assert len(inner.calls) < 6
+
+
+def test_unload_closes_batcher_thread():
+    import time
+    from min_tfs_client_amd.server import ModelManager
+    inner = CountingInner()
+    b = BatchingServable(inner, max_batch_size=4, batch_timeout_s=0.01)
+    mgr = ModelManager()
+    mgr.load("m", b, version=1)
+    assert b._thread.is_alive()
+    mgr.unload("m", 1)
+    b._thread.join(timeout=5)
+    assert not b._thread.is_alive()
