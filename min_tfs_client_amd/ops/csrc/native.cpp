@@ -98,6 +98,10 @@ py::bytes serialize_predict(bool is_request, const std::string& model_name,
     metas[i].shape.assign(sizes.begin(), sizes.end());
     metas[i].content_bytes =
         uint64_t(contig[i].numel()) * contig[i].element_size();
+    // protobuf bytes fields cap at 2GB (TF enforces the same limit)
+    TORCH_CHECK(metas[i].content_bytes < (uint64_t(1) << 31),
+                "tensor '", names[i], "' exceeds the 2GB tensor_content "
+                "limit of the protobuf wire format");
   }
   auto plan = tfswire::plan_predict_message(is_request, model_name, version,
                                             signature, names, metas);
