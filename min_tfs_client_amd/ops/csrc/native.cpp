@@ -274,6 +274,36 @@ py::bytes echo_predict(py::buffer data) {
 
 }  // namespace
 
+// Zero-copy span parse: returns per-tensor (dtype, shape, offset, length)
+// into the SOURCE buffer, no tensor construction — the python side wraps
+// views with torch.frombuffer, borrowing the response bytes' memory.
+py::tuple parse_predict_spans(py::buffer data, bool is_request) {
+  py::buffer_info info = data.request();
+  auto parsed = tfswire::parse_predict_message(
+      static_cast<const uint8_t*>(info.ptr), size_t(info.size), is_request);
+  py::list spans;
+  for (auto& t : parsed.tensors) {
+    if (t.content == nullptr) {
+      spans.append(py::none());  // typed-field tensor: caller falls back
+      continue;
+    }
+    py::dict d;
+    d["name"] = t.name;
+    d["dtype"] = t.dtype;
+    d["shape"] = t.shape;
+    d["offset"] = uint64_t(t.content -
+                           static_cast<const uint8_t*>(info.ptr));
+    d["nbytes"] = t.content_bytes;
+    spans.append(d);
+  }
+  py::dict spec;
+  spec["name"] = parsed.model_spec.name;
+  spec["version"] = parsed.model_spec.version;
+  spec["signature_name"] = parsed.model_spec.signature_name;
+  spec["version_label"] = parsed.model_spec.version_label;
+  return py::make_tuple(spec, spans);
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "MI355X-native TF-Serving codec + CDNA4 pack kernels";
   m.def("serialize_predict_request",
@@ -307,6 +337,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         },
         py::arg("data"), py::arg("device") = "cpu", py::arg("copy_mode") = 0);
   m.def("echo_predict", &echo_predict, py::arg("data"));
+  m.def("parse_predict_spans", &parse_predict_spans, py::arg("data"),
+        py::arg("is_request") = false);
   m.def("tensor_content_bytes",
         [](const at::Tensor& t, int copy_mode) {
           auto c = t.contiguous();

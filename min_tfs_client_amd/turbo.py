@@ -111,10 +111,13 @@ class TurboPredictClient:
                 model_version: Optional[int] = None,
                 signature_name: str = "",
                 output_device: Optional[Union[str, "torch.device"]] = None,
-                copy_mode: int = 1) -> Dict[str, "torch.Tensor"]:
+                copy_mode: int = 1,
+                zero_copy: bool = False) -> Dict[str, "torch.Tensor"]:
         """One Predict round trip. ``output_device``: where response
         tensors land ("cpu" default; "cuda:N" unpacks over the staging
-        pipeline straight to HBM)."""
+        pipeline straight to HBM). ``zero_copy=True`` (CPU outputs only)
+        returns read-only tensor views borrowing the response buffer —
+        no copy at all."""
         with trace_span("turbo.serialize", model=model_name,
                         bytes=sum(t.numel() * t.element_size()
                                   for t in inputs.values())):
@@ -125,10 +128,37 @@ class TurboPredictClient:
             resp = self._predict(blob, timeout)
         self.metrics.observe_bytes("rx", len(resp))
         dev = str(output_device) if output_device is not None else "cpu"
+        if zero_copy and dev == "cpu":
+            with trace_span("turbo.parse", device=dev, zero_copy=True):
+                return self._parse_zero_copy(resp)
         with trace_span("turbo.parse", device=dev):
             _spec, outputs, _ = self._native.parse_predict_response(
                 resp, dev, copy_mode)
         return outputs
+
+    _TF_TO_TORCH = None
+
+    def _parse_zero_copy(self, resp: bytes):
+        from .constants import TF_TO_TORCH_MAPPING
+        if TurboPredictClient._TF_TO_TORCH is None:
+            from .constants import ENUM_TO_TF_MAPPING
+            TurboPredictClient._TF_TO_TORCH = {
+                enum: TF_TO_TORCH_MAPPING[name]
+                for enum, name in ENUM_TO_TF_MAPPING.items()
+                if name in TF_TO_TORCH_MAPPING}
+        _spec, spans = self._native.parse_predict_spans(resp, False)
+        out = {}
+        for d in spans:
+            if d is None:
+                # typed-field tensor: fall back to the copying parse
+                _s, full, _ = self._native.parse_predict_response(
+                    resp, "cpu", 1)
+                return full
+            dtype = TurboPredictClient._TF_TO_TORCH[d["dtype"]]
+            mv = memoryview(resp)[d["offset"]:d["offset"] + d["nbytes"]]
+            t = torch.frombuffer(mv, dtype=dtype)
+            out[d["name"]] = t.reshape(d["shape"])
+        return out
 
     def predict_sharded(self, model_name: str,
                         inputs: Dict[str, "torch.Tensor"],

@@ -213,3 +213,15 @@ def test_output_filter_raw_path(raw_server):
         resp_bytes = c._predict(req.SerializeToString(), 30)
     resp = pb.PredictResponse.FromString(resp_bytes)
     assert sorted(resp.outputs) == ["a"]
+
+
+def test_zero_copy_parse(raw_server):
+    import warnings
+    with TurboPredictClient(raw_server.address) as c:
+        x = torch.randn(8, 16)
+        with warnings.catch_warnings():
+            warnings.simplefilter("ignore")  # frombuffer non-writable note
+            out = c.predict("default", {"x": x}, zero_copy=True)
+        assert torch.equal(out["x"], x)
+        # it's a view over the response buffer, not a copy
+        assert out["x"].data_ptr() != 0
