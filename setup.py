@@ -9,7 +9,7 @@ wire layer needs no codegen, the native build is the codec + HIP kernels).
 """
 import os
 
-from setuptools import setup
+from setuptools import find_namespace_packages, setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
@@ -35,7 +35,17 @@ ext = CUDAExtension(
 setup(
     name="min-tfs-client-amd",
     version="0.1.0",
-    packages=["min_tfs_client_amd"],
+    description="MI355X-native TensorFlow-Serving client + serving "
+                "framework (byte-compatible wire format, HIP pack path)",
+    # the tensorflow/tensorflow_serving pb2 shims and the min_tfs_client
+    # alias are namespace packages, like the reference wheel's generated
+    # modules (reference setup.py:100-101 find_namespace_packages)
+    packages=find_namespace_packages(
+        include=["min_tfs_client_amd", "min_tfs_client_amd.*",
+                 "min_tfs_client", "tensorflow", "tensorflow.*",
+                 "tensorflow_serving", "tensorflow_serving.*"]),
     ext_modules=[ext],
+    install_requires=["numpy", "grpcio>=1.21", "protobuf>=3.8", "torch"],
+    python_requires=">=3.10",
     cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
 )
