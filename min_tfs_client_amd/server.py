@@ -289,6 +289,12 @@ class PredictionServiceImpl(PredictionServiceServicer):
                    f"Expected a classification signature for model "
                    f"{request.model_spec.name}")
         result = classify_fn(request.input)
+        # example-count metric (servables/tensorflow/util.cc:36-66 analogue)
+        n_examples = len(request.input.example_list.examples) or \
+            len(request.input.example_list_with_context.examples)
+        self.metrics.inc(
+            f"request_example_counts{{model={request.model_spec.name!r}}}",
+            n_examples)
         response = pb.ClassificationResponse()
         response.model_spec.CopyFrom(request.model_spec)
         response.result.CopyFrom(result)
@@ -304,6 +310,11 @@ class PredictionServiceImpl(PredictionServiceServicer):
                    f"Expected a regression signature for model "
                    f"{request.model_spec.name}")
         result = regress_fn(request.input)
+        n_examples = len(request.input.example_list.examples) or \
+            len(request.input.example_list_with_context.examples)
+        self.metrics.inc(
+            f"request_example_counts{{model={request.model_spec.name!r}}}",
+            n_examples)
         response = pb.RegressionResponse()
         response.model_spec.CopyFrom(request.model_spec)
         response.result.CopyFrom(result)
