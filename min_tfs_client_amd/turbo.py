@@ -139,6 +139,8 @@ class TurboPredictClient:
     _TF_TO_TORCH = None
 
     def _parse_zero_copy(self, resp: bytes):
+        import warnings
+
         from .constants import TF_TO_TORCH_MAPPING
         if TurboPredictClient._TF_TO_TORCH is None:
             from .constants import ENUM_TO_TF_MAPPING
@@ -156,7 +158,11 @@ class TurboPredictClient:
                 return full
             dtype = TurboPredictClient._TF_TO_TORCH[d["dtype"]]
             mv = memoryview(resp)[d["offset"]:d["offset"] + d["nbytes"]]
-            t = torch.frombuffer(mv, dtype=dtype)
+            with warnings.catch_warnings():
+                # frombuffer warns that bytes are read-only: that is the
+                # documented contract of zero_copy=True
+                warnings.simplefilter("ignore")
+                t = torch.frombuffer(mv, dtype=dtype)
             out[d["name"]] = t.reshape(d["shape"])
         return out
 
