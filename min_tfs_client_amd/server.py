@@ -104,7 +104,9 @@ class _Version:
 
 
 class ModelManager:
-    """Servable registry with TF-Serving-style version states."""
+    """Servable registry with TF-Serving-style version states and a state
+    event bus (ServableStateMonitor analogue, core/servable_state_monitor.h:
+    subscribers get (model, version, state) on every transition)."""
 
     def __init__(self):
         self._lock = threading.RLock()
@@ -112,6 +114,37 @@ class ModelManager:
         # {model: {label: version}} — ModelSpec.version_label routing
         # (model.proto:9-33 oneof version_choice; ModelConfig.version_labels)
         self._labels: Dict[str, Dict[str, int]] = {}
+        self._subscribers = []
+
+    # -- state event bus ------------------------------------------------
+    def subscribe(self, callback) -> None:
+        """callback(model_name, version, state) on every transition."""
+        with self._lock:
+            self._subscribers.append(callback)
+
+    def _notify(self, name: str, version: int, state: int) -> None:
+        for cb in list(self._subscribers):
+            try:
+                cb(name, version, state)
+            except Exception:  # noqa: BLE001 - observers must not break serving
+                pass
+
+    def wait_for_state(self, name: str, version: int, state: int,
+                       timeout: float = 30.0) -> bool:
+        """Blocks until (name, version) reaches `state` (monitor-style
+        helper used by tests and warm-start orchestration)."""
+        event = threading.Event()
+
+        def cb(n, v, s):
+            if n == name and v == version and s == state:
+                event.set()
+
+        self.subscribe(cb)
+        with self._lock:
+            v = self._models.get(name, {}).get(version)
+            if v is not None and v.state == state:
+                return True
+        return event.wait(timeout)
 
     # -- lifecycle ------------------------------------------------------
     def load(self, name: str, servable: Servable, version: int = 1) -> None:
@@ -120,7 +153,9 @@ class ModelManager:
             v = _Version(servable)
             versions[version] = v
             v.state = STATE_LOADING
+            self._notify(name, version, STATE_LOADING)
             v.state = STATE_AVAILABLE
+        self._notify(name, version, STATE_AVAILABLE)
 
     def fail_load(self, name: str, version: int, code: int, msg: str) -> None:
         with self._lock:
@@ -129,6 +164,7 @@ class ModelManager:
             v.state = STATE_END
             v.status_error = (code, msg)
             versions[version] = v
+        self._notify(name, version, STATE_END)
 
     def unload(self, name: str, version: Optional[int] = None) -> None:
         with self._lock:
@@ -139,12 +175,14 @@ class ModelManager:
             for ver in targets:
                 if ver in versions:
                     versions[ver].state = STATE_UNLOADING
+                    self._notify(name, ver, STATE_UNLOADING)
                     servable = versions[ver].servable
                     close = getattr(servable, "close", None)
                     if callable(close):
                         close()  # e.g. BatchingServable's batcher thread
                     versions[ver].state = STATE_END
                     versions[ver].servable = None
+                    self._notify(name, ver, STATE_END)
 
     def set_version_label(self, name: str, label: str,
                           version: int) -> None:

@@ -200,3 +200,21 @@ def test_version_label_routing(server):
             server.manager.set_version_label("labeled", "bad", 99)
     finally:
         c.close()
+
+
+def test_state_event_bus():
+    """ServableStateMonitor analogue: subscribers see every transition,
+    wait_for_state blocks until a target state."""
+    from min_tfs_client_amd.server import ModelManager
+    mgr = ModelManager()
+    events = []
+    mgr.subscribe(lambda n, v, s: events.append((n, v, s)))
+    mgr.load("m", identity_servable(), version=1)
+    assert (("m", 1, 20) in events) and (("m", 1, 30) in events)
+    assert mgr.wait_for_state("m", 1, 30, timeout=1)
+    mgr.unload("m", 1)
+    assert events[-1] == ("m", 1, 50)
+    # a broken subscriber must not break serving
+    mgr.subscribe(lambda *a: 1 / 0)
+    mgr.load("m2", identity_servable(), version=1)
+    assert mgr.get("m2") is not None
