@@ -125,6 +125,18 @@ int test_typed_field_parse() {
   return 0;
 }
 
+int test_version_label_parse() {
+  // ModelSpec{name="m", version_label="prod"(field 4)}
+  std::vector<uint8_t> msg = {0x0a, 0x0b, 0x0a, 0x01, 'm',
+                              0x22, 0x04, 'p', 'r', 'o', 'd', 0x00};
+  msg[1] = 0x09;  // spec length: name(3) + label(6)
+  msg.resize(11);
+  auto parsed = parse_predict_message(msg.data(), msg.size(), true);
+  CHECK(parsed.model_spec.name == "m");
+  CHECK(parsed.model_spec.version_label == "prod");
+  return 0;
+}
+
 int test_unknown_fields_skipped() {
   // field 99 varint, then a valid model_spec
   std::vector<uint8_t> msg = {0xd8, 0x06, 0x07,       // field 99 = 7
@@ -141,6 +153,7 @@ int main() {
   rc |= test_response_field_numbers();
   rc |= test_truncated_inputs_do_not_overrun();
   rc |= test_typed_field_parse();
+  rc |= test_version_label_parse();
   rc |= test_unknown_fields_skipped();
   if (rc == 0) std::printf("wire_test: %d checks passed\n", tests_run);
   return rc;
