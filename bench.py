@@ -105,6 +105,11 @@ def main():
                          "many parallel rpcs over separate channels "
                          "(measured: 4 shards more than halve p50 RTT on "
                          "the 19MB config; 1 disables)")
+    ap.add_argument("--servers", type=int, default=1,
+                    help="loopback server processes per rank; the python "
+                         "gRPC server caps ~9 GB/s per process, so >1 "
+                         "raises the per-GPU ceiling (requests spread "
+                         "across instances)")
     ap.add_argument("--pipeline", type=int, default=1,
                     help="in-flight requests per rank (1 = sequential; "
                          ">1 overlaps serialize/transport/parse of "
@@ -130,30 +135,36 @@ def main():
         backend = "nccl" if has_gpu else "gloo"
         dist.init_process_group(backend=backend)
 
-    # ---- per-rank loopback server ------------------------------------
-    if args.transport == "unix":
-        address = f"unix:///tmp/mi355x_bench_{os.getpid()}_{rank}.sock"
-    else:
-        address = None
+    # ---- per-rank loopback server fleet -------------------------------
     ctx = multiprocessing.get_context("spawn")
-    ready, stop = ctx.Event(), ctx.Event()
-    if address is None:
-        # tcp: pick port in parent to pass to child
-        import socket as _socket
-        s = _socket.socket()
-        s.bind(("127.0.0.1", 0))
-        port = s.getsockname()[1]
-        s.close()
-        address = f"127.0.0.1:{port}"
     server_device = f"cuda:{local_rank}" if has_gpu else "cpu"
-    proc = ctx.Process(target=_server_proc,
-                       args=(address, ready, stop, args.servable,
-                             server_device),
-                       daemon=True)
-    proc.start()
-    # fresh boxes: first torch import in the child can take minutes
-    if not ready.wait(300):
-        raise RuntimeError("bench server failed to start")
+    addresses, procs, stop = [], [], ctx.Event()
+    readies = []
+    for si in range(max(1, args.servers)):
+        if args.transport == "unix":
+            address = (f"unix:///tmp/mi355x_bench_{os.getpid()}_"
+                       f"{rank}_{si}.sock")
+        else:
+            import socket as _socket
+            s = _socket.socket()
+            s.bind(("127.0.0.1", 0))
+            port = s.getsockname()[1]
+            s.close()
+            address = f"127.0.0.1:{port}"
+        ready = ctx.Event()
+        proc = ctx.Process(target=_server_proc,
+                           args=(address, ready, stop, args.servable,
+                                 server_device),
+                           daemon=True)
+        proc.start()
+        addresses.append(address)
+        readies.append(ready)
+        procs.append(proc)
+    # fresh boxes: first torch import in the children can take minutes
+    for ready in readies:
+        if not ready.wait(300):
+            raise RuntimeError("bench server failed to start")
+    address = addresses[0] if len(addresses) == 1 else addresses
 
     # ---- client ------------------------------------------------------
     inputs = make_inputs(args.bench_config, device)
@@ -164,7 +175,8 @@ def main():
         from min_tfs_client_amd.turbo import TurboPredictClient
         client = TurboPredictClient(
             address,
-            num_channels=max(min(args.pipeline, 8), args.shards))
+            num_channels=max(min(args.pipeline, 8), args.shards,
+                             args.servers))
 
         def step_fn(step_inputs):
             out_dev = f"cuda:{local_rank}" if has_gpu else "cpu"
@@ -177,10 +189,10 @@ def main():
                                   copy_mode=args.copy_mode)
     else:
         from min_tfs_client_amd.client import TensorServingClient
-        host, port = address.split("//")[-1], None
         if args.transport == "unix":
             raise SystemExit("--encoding proto requires --transport tcp")
-        host, port = address.split(":")
+        addr0 = address if isinstance(address, str) else address[0]
+        host, port = addr0.split(":")
         client = TensorServingClient(host, int(port))
 
         def step_fn(step_inputs):
@@ -329,13 +341,15 @@ def main():
                 "encoding": args.encoding,
                 "copy_mode": args.copy_mode,
                 "transport": args.transport,
+                "servers_per_rank": args.servers,
                 "gpu": has_gpu,
             },
         }
         print(json.dumps(result))
 
     stop.set()
-    proc.join(timeout=10)
+    for proc in procs:
+        proc.join(timeout=10)
     if dist is not None:
         dist.destroy_process_group()
 

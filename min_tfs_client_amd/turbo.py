@@ -49,27 +49,34 @@ class TurboPredictClient:
     """Raw-bytes Predict client over the C++ codec.
 
     ``target``: "host:port" or "unix:///path.sock" (unix sockets cut
-    loopback syscall overhead — preferred for same-host serving).
+    loopback syscall overhead — preferred for same-host serving), or a
+    LIST of targets — a local serving fleet; requests round-robin across
+    instances and ``predict_sharded`` parallelizes one request over all
+    of them (a python gRPC server process caps at ~9 GB/s, so multiple
+    instances per GPU raise the ceiling — profiles/).
     """
 
-    def __init__(self, target: str,
+    def __init__(self, target,
                  credentials: Optional[grpc.ChannelCredentials] = None,
                  options: Optional[list] = None,
                  num_channels: int = 1):
         self._native = require_native()
         tune_malloc()  # large wire buffers: arena reuse, no per-call mmap
         opts = _CHANNEL_OPTS + (options or [])
+        targets = [target] if isinstance(target, str) else list(target)
         self._channels = []
         self._stubs = []
-        for i in range(max(1, num_channels)):
+        n = max(len(targets), max(1, num_channels))
+        for i in range(n):
             # separate HTTP/2 connections (no shared subchannel) ->
             # parallel transport for pipelined requests
             copts = opts + [("grpc.use_local_subchannel_pool", 1),
                             ("grpc.channel_id", i)]
+            tgt = targets[i % len(targets)]
             if credentials:
-                ch = grpc.secure_channel(target, credentials, options=copts)
+                ch = grpc.secure_channel(tgt, credentials, options=copts)
             else:
-                ch = grpc.insecure_channel(target, options=copts)
+                ch = grpc.insecure_channel(tgt, options=copts)
             self._channels.append(ch)
             self._stubs.append(ch.unary_unary(
                 _PREDICT_PATH, request_serializer=_identity,

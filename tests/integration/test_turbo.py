@@ -225,3 +225,22 @@ def test_zero_copy_parse(raw_server):
         assert torch.equal(out["x"], x)
         # it's a view over the response buffer, not a copy
         assert out["x"].data_ptr() != 0
+
+
+def test_multi_target_client():
+    """A client over a fleet of server instances: round-robin +
+    shard-parallel across targets."""
+    with ModelServer(port=0, raw_predict=True) as s1, \
+            ModelServer(port=0, raw_predict=True) as s2:
+        s1.manager.load("m", identity_servable(), version=1)
+        s2.manager.load("m", identity_servable(), version=1)
+        with TurboPredictClient([s1.address, s2.address]) as c:
+            assert len(c._stubs) == 2
+            x = torch.randn(8, 4)
+            for _ in range(3):
+                out = c.predict_sharded("m", {"x": x}, shards=2)
+                assert torch.equal(out["x"], x)
+            # both servers actually served requests
+            n1 = s1.metrics.latency_quantiles("predict").get("count", 0)
+            n2 = s2.metrics.latency_quantiles("predict").get("count", 0)
+            assert n1 > 0 and n2 > 0
