@@ -34,21 +34,33 @@ def _client_proc(address, n_requests, pipeline, q):
         q.put((done, time.perf_counter() - t0))
 
 
-def main(n_procs=3, per_proc=150, pipeline=8):
+def main(n_procs=3, per_proc=150, pipeline=8, pair_servers=0):
     from min_tfs_client_amd.server import ModelServer, identity_servable
-    sock = f"unix:///tmp/mi355x_mp_{os.getpid()}.sock"
     ctx = multiprocessing.get_context("spawn")
-    with ModelServer(address=sock, raw_predict=True, max_workers=32) as srv:
+    servers = []
+    socks = []
+    n_servers = n_procs if pair_servers else 1
+    for i in range(n_servers):
+        sock = f"unix:///tmp/mi355x_mp_{os.getpid()}_{i}.sock"
+        srv = ModelServer(address=sock, raw_predict=True, max_workers=32)
         srv.manager.load("m", identity_servable(), version=1)
+        srv.start()
+        servers.append(srv)
+        socks.append(sock)
+    try:
         q = ctx.Queue()
         procs = [ctx.Process(target=_client_proc,
-                             args=(sock, per_proc, pipeline, q))
-                 for _ in range(n_procs)]
+                             args=(socks[i % len(socks)], per_proc,
+                                   pipeline, q))
+                 for i in range(n_procs)]
         for p in procs:
             p.start()
         results = [q.get(timeout=600) for _ in procs]
         for p in procs:
             p.join(timeout=30)
+    finally:
+        for srv in servers:
+            srv.stop(0)
     total = sum(r[0] for r in results)
     # clients start within ~a second of each other (spawn+import outside
     # their timed regions); the slowest child's request-loop time is the
@@ -56,6 +68,7 @@ def main(n_procs=3, per_proc=150, pipeline=8):
     window = max(r[1] for r in results)
     print(json.dumps({
         "client_procs": n_procs, "pipeline": pipeline,
+        "paired_servers": bool(pair_servers),
         "total_requests": total,
         "window_s": round(window, 2),
         "agg_req_per_s": round(total / window, 1),
