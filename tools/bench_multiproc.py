@@ -77,4 +77,4 @@ def main(n_procs=3, per_proc=150, pipeline=8, pair_servers=0):
 
 
 if __name__ == "__main__":
-    main(*(int(a) for a in sys.argv[1:4]))
+    main(*(int(a) for a in sys.argv[1:5]))
