@@ -38,13 +38,13 @@ import torch  # noqa: E402
 
 
 def _server_proc(address: str, ready, stop, servable: str = "echo",
-                 device: str = "cpu"):
+                 device: str = "cpu", shm_dir: str = ""):
     """Loopback PredictionService in its own process (own GIL)."""
     if _ROOT not in sys.path:
         sys.path.insert(0, _ROOT)
     from min_tfs_client_amd.server import ModelServer, identity_servable
     with ModelServer(address=address, raw_predict=True, max_workers=16,
-                     device=device) as srv:
+                     device=device, shm_handshake_dir=shm_dir or None) as srv:
         if servable == "echo":
             srv.manager.load("default", identity_servable(), version=1)
         elif servable == "resnet50":
@@ -95,7 +95,10 @@ def main():
     ap.add_argument("--copy-mode", type=int, default=1,
                     help="0 = pinned-staged pipelined copies, 1 = direct "
                          "pageable hipMemcpy (A/B)")
-    ap.add_argument("--transport", default="unix", choices=["unix", "tcp"])
+    ap.add_argument("--transport", default="unix",
+                    choices=["unix", "tcp", "shm"],
+                    help="unix/tcp = gRPC; shm = shared-memory local "
+                         "transport (same wire bytes, ~2 copies per hop)")
     ap.add_argument("--servable", default="echo",
                     choices=["echo", "resnet50", "bert"],
                     help="what the loopback server runs: echo (the codec/"
@@ -140,8 +143,13 @@ def main():
     server_device = f"cuda:{local_rank}" if has_gpu else "cpu"
     addresses, procs, stop = [], [], ctx.Event()
     readies = []
+    shm_dirs = []
     for si in range(max(1, args.servers)):
-        if args.transport == "unix":
+        shm_dir = ""
+        if args.transport == "shm":
+            shm_dir = f"/tmp/mi355x_shm_{os.getpid()}_{rank}_{si}"
+            shm_dirs.append(shm_dir)
+        if args.transport in ("unix", "shm"):
             address = (f"unix:///tmp/mi355x_bench_{os.getpid()}_"
                        f"{rank}_{si}.sock")
         else:
@@ -154,7 +162,7 @@ def main():
         ready = ctx.Event()
         proc = ctx.Process(target=_server_proc,
                            args=(address, ready, stop, args.servable,
-                                 server_device),
+                                 server_device, shm_dir),
                            daemon=True)
         proc.start()
         addresses.append(address)
@@ -171,7 +179,57 @@ def main():
     per_rank_batch = next(iter(inputs.values())).shape[0] \
         if next(iter(inputs.values())).dim() > 0 else 1
 
-    if args.encoding == "turbo":
+    if args.transport == "shm":
+        if args.encoding != "turbo" or args.pipeline > 1:
+            raise SystemExit("--transport shm requires turbo encoding and "
+                             "no pipelining (open more connections "
+                             "instead)")
+        from min_tfs_client_amd.shm import ShmPredictClient
+
+        class _ShmFleet:
+            """shards parallel connections over the server fleet."""
+
+            def __init__(self, dirs, conns):
+                from concurrent.futures import ThreadPoolExecutor
+                self.clients = [
+                    ShmPredictClient(dirs[i % len(dirs)],
+                                     slot_bytes=96 << 20)
+                    for i in range(conns)]
+                self.pool = ThreadPoolExecutor(max_workers=conns)
+
+            def close(self):
+                for c in self.clients:
+                    c.close()
+
+        fleet = _ShmFleet(shm_dirs, max(1, args.shards))
+        client = fleet.clients[0]
+
+        def step_fn(step_inputs):
+            out_dev = f"cuda:{local_rank}" if has_gpu else "cpu"
+            ncl = len(fleet.clients)
+            keys = list(step_inputs.keys())
+            batch = (step_inputs[keys[0]].shape[0]
+                     if step_inputs[keys[0]].dim() > 0 else 1)
+            if ncl == 1 or batch < ncl or any(
+                    step_inputs[k].dim() == 0 or
+                    step_inputs[k].shape[0] != batch for k in keys):
+                return client.predict("default", step_inputs,
+                                      output_device=out_dev,
+                                      copy_mode=args.copy_mode)
+            base, rem = divmod(batch, ncl)
+            sizes = [base + (1 if i < rem else 0) for i in range(ncl)]
+            futs, off = [], 0
+            for i, nrows in enumerate(sizes):
+                shard = {k: step_inputs[k].narrow(0, off, nrows)
+                         for k in keys}
+                futs.append(fleet.pool.submit(
+                    fleet.clients[i].predict, "default", shard,
+                    output_device=out_dev, copy_mode=args.copy_mode))
+                off += nrows
+            parts = [f.result() for f in futs]
+            return {k: torch.cat([p[k] for p in parts], dim=0)
+                    for k in parts[0]}
+    elif args.encoding == "turbo":
         from min_tfs_client_amd.turbo import TurboPredictClient
         client = TurboPredictClient(
             address,

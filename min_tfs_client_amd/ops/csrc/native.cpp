@@ -9,8 +9,11 @@
 
 #include <torch/extension.h>
 
+#include <atomic>
+#include <chrono>
 #include <cstring>
 #include <string>
+#include <thread>
 #include <vector>
 
 #include "wire.h"
@@ -274,6 +277,90 @@ py::bytes echo_predict(py::buffer data) {
 
 }  // namespace
 
+// serialize into a caller-provided buffer (shared-memory transport):
+// returns bytes written; throws if capacity is too small.
+uint64_t serialize_predict_into(py::buffer dst, bool is_request,
+                                const std::string& model_name,
+                                int64_t version,
+                                const std::string& signature,
+                                const std::vector<std::string>& names,
+                                const std::vector<at::Tensor>& tensors,
+                                int copy_mode) {
+  TORCH_CHECK(names.size() == tensors.size(), "names/tensors mismatch");
+  py::buffer_info info = dst.request(true);
+  std::vector<tfswire::TensorMeta> metas(tensors.size());
+  std::vector<at::Tensor> contig(tensors.size());
+  for (size_t i = 0; i < tensors.size(); ++i) {
+    contig[i] = tensors[i].contiguous();
+    metas[i].dtype = torch_to_tf_dtype(contig[i].scalar_type());
+    auto sizes = contig[i].sizes();
+    metas[i].shape.assign(sizes.begin(), sizes.end());
+    metas[i].content_bytes =
+        uint64_t(contig[i].numel()) * contig[i].element_size();
+    TORCH_CHECK(metas[i].content_bytes < (uint64_t(1) << 31),
+                "tensor exceeds the 2GB tensor_content limit");
+  }
+  auto plan = tfswire::plan_predict_message(is_request, model_name, version,
+                                            signature, names, metas);
+  TORCH_CHECK(plan.total_size <= uint64_t(info.size),
+              "shm slot too small: need ", plan.total_size, " bytes, have ",
+              info.size);
+  auto* buf = static_cast<uint8_t*>(info.ptr);
+  tfswire::write_predict_message(buf, plan, is_request, model_name, version,
+                                 signature, names, metas);
+  {
+    py::gil_scoped_release release;
+    for (size_t i = 0; i < contig.size(); ++i) {
+      const auto& span = plan.spans[i];
+      if (span.nbytes == 0) continue;
+      if (contig[i].is_cuda()) {
+        mi355x::copy_device_to_host_ptr(contig[i], buf + span.offset,
+                                        span.nbytes, copy_mode);
+      } else {
+        std::memcpy(buf + span.offset, contig[i].const_data_ptr(),
+                    span.nbytes);
+      }
+    }
+  }
+  return plan.total_size;
+}
+
+// GIL-released wait on a shared u32 (x86 cross-process visibility):
+// returns the observed value, or 0xFFFFFFFF on timeout.
+uint32_t shm_wait_value(py::buffer buf, uint64_t offset, uint32_t target,
+                        double timeout_s) {
+  py::buffer_info info = buf.request();
+  auto* p = reinterpret_cast<volatile uint32_t*>(
+      static_cast<uint8_t*>(info.ptr) + offset);
+  py::gil_scoped_release release;
+  const auto deadline = std::chrono::steady_clock::now() +
+      std::chrono::duration<double>(timeout_s);
+  int spins = 0;
+  while (true) {
+    uint32_t v = *p;
+    if (v == target) {
+      std::atomic_thread_fence(std::memory_order_acquire);
+      return v;
+    }
+    if (std::chrono::steady_clock::now() > deadline) return 0xFFFFFFFFu;
+    if (++spins < 1024) {
+#if defined(__x86_64__)
+      __builtin_ia32_pause();
+#endif
+    } else {
+      std::this_thread::sleep_for(std::chrono::microseconds(
+          spins < 4096 ? 5 : 50));
+    }
+  }
+}
+
+void shm_store_value(py::buffer buf, uint64_t offset, uint32_t value) {
+  py::buffer_info info = buf.request(true);
+  auto* p = reinterpret_cast<std::atomic<uint32_t>*>(
+      static_cast<uint8_t*>(info.ptr) + offset);
+  p->store(value, std::memory_order_release);
+}
+
 // Zero-copy span parse: returns per-tensor (dtype, shape, offset, length)
 // into the SOURCE buffer, no tensor construction — the python side wraps
 // views with torch.frombuffer, borrowing the response bytes' memory.
@@ -339,6 +426,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("echo_predict", &echo_predict, py::arg("data"));
   m.def("parse_predict_spans", &parse_predict_spans, py::arg("data"),
         py::arg("is_request") = false);
+  m.def("serialize_predict_into", &serialize_predict_into,
+        py::arg("dst"), py::arg("is_request"), py::arg("model_name"),
+        py::arg("version"), py::arg("signature"), py::arg("names"),
+        py::arg("tensors"), py::arg("copy_mode") = 1);
+  m.def("shm_wait_value", &shm_wait_value, py::arg("buf"),
+        py::arg("offset"), py::arg("target"), py::arg("timeout_s"));
+  m.def("shm_store_value", &shm_store_value, py::arg("buf"),
+        py::arg("offset"), py::arg("value"));
   m.def("tensor_content_bytes",
         [](const at::Tensor& t, int copy_mode) {
           auto c = t.contiguous();

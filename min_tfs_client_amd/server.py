@@ -553,7 +553,8 @@ class ModelServer:
                  servable_factory=None,
                  raw_predict: bool = False,
                  device: str = "cpu",
-                 address: Optional[str] = None):
+                 address: Optional[str] = None,
+                 shm_handshake_dir: Optional[str] = None):
         from .utils.allocator import tune_malloc
         tune_malloc()
         from .request_logging import ServerRequestLogger
@@ -609,6 +610,12 @@ class ModelServer:
                                                     self._server)
             add_ModelServiceServicer_to_server(self.model_service,
                                                self._server)
+        self.shm_listener = None
+        if shm_handshake_dir:
+            from .shm import ShmListener
+            self.shm_listener = ShmListener(self.manager,
+                                            shm_handshake_dir,
+                                            device=device)
         if address is not None:
             self.address = address
             self.port = self._server.add_insecure_port(address)
@@ -618,9 +625,13 @@ class ModelServer:
 
     def start(self) -> "ModelServer":
         self._server.start()
+        if self.shm_listener is not None:
+            self.shm_listener.start()
         return self
 
     def stop(self, grace: Optional[float] = None) -> None:
+        if self.shm_listener is not None:
+            self.shm_listener.stop()
         self._server.stop(grace)
 
     def __enter__(self):
