@@ -103,11 +103,13 @@ def main():
                     choices=["echo", "resnet50", "bert"],
                     help="what the loopback server runs: echo (the codec/"
                          "transport benchmark) or a real model family")
-    ap.add_argument("--shards", type=int, default=4,
+    ap.add_argument("--shards", type=int, default=-1,
                     help="split each logical request along dim 0 into this "
                          "many parallel rpcs over separate channels "
-                         "(measured: 4 shards more than halve p50 RTT on "
-                         "the 19MB config; 1 disables)")
+                         "(default: 4 for gRPC transports — measured to "
+                         "more than halve p50 — and 1 for shm, where the "
+                         "copies are the only cost and sharding adds "
+                         "thread overhead)")
     ap.add_argument("--servers", type=int, default=1,
                     help="loopback server processes per rank; the python "
                          "gRPC server caps ~9 GB/s per process, so >1 "
@@ -118,6 +120,8 @@ def main():
                          ">1 overlaps serialize/transport/parse of "
                          "consecutive requests)")
     args = ap.parse_args()
+    if args.shards < 0:
+        args.shards = 1 if args.transport == "shm" else 4
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
