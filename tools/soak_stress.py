@@ -16,6 +16,7 @@ import torch  # noqa: E402
 from min_tfs_client_amd.batching import BatchingServable  # noqa: E402
 from min_tfs_client_amd.models import resnet50_servable  # noqa: E402
 from min_tfs_client_amd.server import ModelServer, Servable, identity_servable  # noqa: E402
+from min_tfs_client_amd.shm import ShmPredictClient  # noqa: E402
 from min_tfs_client_amd.turbo import TurboPredictClient  # noqa: E402
 
 
@@ -37,8 +38,9 @@ def main(seconds=300):
     def scaled(inputs):
         return {k: v * 2 for k, v in inputs.items()}
 
+    shm_dir = f"/tmp/mi355x_stress_hs_{os.getpid()}"
     with ModelServer(address=sock, raw_predict=True, device=dev,
-                     max_workers=32) as srv:
+                     max_workers=32, shm_handshake_dir=shm_dir) as srv:
         srv.manager.load("echo", identity_servable(), version=1)
         srv.manager.load("scale", Servable(scaled), version=1)
         if dev != "cpu":
@@ -100,6 +102,18 @@ def main(seconds=300):
                     except Exception as e:  # noqa: BLE001
                         record("model", False, e)
 
+        def shm_worker(wid):
+            with ShmPredictClient(shm_dir, slot_bytes=64 << 20) as c:
+                g = torch.Generator().manual_seed(1000 + wid)
+                while time.monotonic() < stop_at:
+                    x = torch.randn(16, 3, 96, 96, generator=g).to(dev)
+                    try:
+                        out = c.predict("scale", {"x": x},
+                                        output_device=dev, timeout=60)
+                        record("shm", torch.allclose(out["x"], x * 2))
+                    except Exception as e:  # noqa: BLE001
+                        record("shm", False, e)
+
         def zc_worker():
             with TurboPredictClient(sock) as c:
                 x = torch.randn(4, 256)
@@ -114,6 +128,8 @@ def main(seconds=300):
                             record("zerocopy", False, e)
 
         workers = ([threading.Thread(target=seq_worker, args=(i,))
+                    for i in range(2)] +
+                   [threading.Thread(target=shm_worker, args=(i,))
                     for i in range(2)] +
                    [threading.Thread(target=pipe_worker),
                     threading.Thread(target=shard_worker),
