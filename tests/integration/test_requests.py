@@ -218,3 +218,46 @@ def test_state_event_bus():
     mgr.subscribe(lambda *a: 1 / 0)
     mgr.load("m2", identity_servable(), version=1)
     assert mgr.get("m2") is not None
+
+
+def test_typed_output_encoding_mode():
+    """output_encoding='typed' reproduces TF-Serving's default
+    AsProtoField responses (predict_util.cc:222-226) — the representation
+    the reference client's decoder requires."""
+    with ModelServer(port=0, output_encoding="typed") as srv:
+        srv.manager.load("m", identity_servable(), version=1)
+        c = TensorServingClient("127.0.0.1", srv.port)
+        try:
+            x = np.array([1.5, 2.5], dtype=np.float32)
+            resp = c.predict_request("m", {"x": x})
+            proto = resp.outputs["x"]
+            assert len(proto.tensor_content) == 0
+            assert list(proto.float_val) == [1.5, 2.5]
+            np.testing.assert_array_equal(tensor_proto_to_ndarray(proto), x)
+        finally:
+            c.close()
+
+
+def test_metadata_signature_unpack():
+    """GetModelMetadata's Any payload unpacks into a SignatureDefMap with
+    the servable's declared inputs/outputs."""
+    from min_tfs_client_amd.models import resnet50_servable
+    from min_tfs_client_amd.wire import messages as pb
+    with ModelServer(port=0) as srv:
+        srv.manager.load("resnet50", resnet50_servable(), version=1)
+        c = TensorServingClient("127.0.0.1", srv.port)
+        try:
+            resp = c.get_model_metadata_request("resnet50")
+            sdm = pb.SignatureDefMap()
+            assert resp.metadata["signature_def"].type_url.endswith(
+                "tensorflow.serving.SignatureDefMap")
+            sdm.MergeFromString(resp.metadata["signature_def"].value)
+            sig = sdm.signature_def["serving_default"]
+            assert sig.method_name == "tensorflow/serving/predict"
+            assert sig.inputs["images"].dtype == 1
+            assert [d.size for d in
+                    sig.inputs["images"].tensor_shape.dim] == \
+                [-1, 3, 224, 224]
+            assert sig.outputs["logits"].name == "logits:0"
+        finally:
+            c.close()

@@ -219,3 +219,42 @@ def test_rest_classify_and_regress(manager):
             _post(rest_srv, "/v1/models/default:classify",
                   {"examples": [{"x": [1.0]}]})
         assert err.value.code == 400
+
+
+def test_cli_monitoring_config(tmp_path):
+    """--monitoring_config_file moves the Prometheus path
+    (monitoring_config.proto:7-19 semantics)."""
+    from min_tfs_client_amd.model_server import build_arg_parser, make_server
+    vdir = tmp_path / "m" / "1"
+    vdir.mkdir(parents=True)
+    (vdir / "identity").touch()
+    mon = tmp_path / "monitoring.config"
+    mon.write_text(
+        'prometheus_config { enable: true path: "/custom/metrics" }')
+    import socket
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    rest_port = s.getsockname()[1]
+    s.close()
+    args = build_arg_parser().parse_args([
+        "--port", "0", "--rest_api_port", str(rest_port),
+        "--model_name", "m", "--model_base_path", str(tmp_path / "m"),
+        "--monitoring_config_file", str(mon),
+        "--file_system_poll_wait_seconds", "0",
+    ])
+    server, source, rest_srv = make_server(args)
+    try:
+        server.start()
+        source.start()
+        rest_srv.start()
+        body = urllib.request.urlopen(
+            f"http://127.0.0.1:{rest_srv.port}/custom/metrics").read()
+        assert b":tensorflow:serving:" in body or body == b"\n"
+        with pytest.raises(urllib.error.HTTPError):
+            urllib.request.urlopen(
+                f"http://127.0.0.1:{rest_srv.port}"
+                f"/monitoring/prometheus/metrics")
+    finally:
+        source.stop()
+        rest_srv.stop()
+        server.stop(0)
