@@ -32,7 +32,7 @@ from .utils.metrics import MetricsRegistry
 
 _MODEL_RE = re.compile(
     r"^/v1/models/(?P<model>[^/:]+)"
-    r"(?:/versions/(?P<version>\d+))?"
+    r"(?:/versions/(?P<version>\d+)|/labels/(?P<label>[^/:]+))?"
     r"(?P<rest>:predict|:classify|:regress|/metadata)?$")
 
 
@@ -218,6 +218,7 @@ class RestApiServer:
                     return
                 name = m.group("model")
                 version = m.group("version")
+                label = m.group("label")
                 method = m.group("rest")
                 length = int(self.headers.get("Content-Length", 0))
                 try:
@@ -227,7 +228,7 @@ class RestApiServer:
                     return
                 try:
                     servable = outer.manager.get(
-                        name, int(version) if version else None)
+                        name, int(version) if version else None, label)
                 except KeyError as e:
                     self._error(404, str(e))
                     return

@@ -258,3 +258,15 @@ def test_cli_monitoring_config(tmp_path):
         source.stop()
         rest_srv.stop()
         server.stop(0)
+
+
+def test_rest_label_route(manager):
+    manager.set_version_label("double", "stable", 3)
+    with RestApiServer(manager, port=0) as rest_srv:
+        body = _post(rest_srv, "/v1/models/double/labels/stable:predict",
+                     {"instances": [[2.0]]})
+        assert body == {"predictions": [[4.0]]}
+        with pytest.raises(urllib.error.HTTPError) as err:
+            _post(rest_srv, "/v1/models/double/labels/nope:predict",
+                  {"instances": [[2.0]]})
+        assert err.value.code == 404
