@@ -220,5 +220,11 @@ class TensorServingClient:
 
 def decode_predict_response(response,
                             as_numpy: bool = True) -> Dict[str, Any]:
-    """Convenience: PredictResponse -> {name: ndarray}."""
-    return {k: tensor_proto_to_ndarray(v) for k, v in response.outputs.items()}
+    """Convenience: PredictResponse -> {name: ndarray} (or torch tensors
+    with as_numpy=False — bf16 outputs then come back as torch.bfloat16)."""
+    if as_numpy:
+        return {k: tensor_proto_to_ndarray(v)
+                for k, v in response.outputs.items()}
+    from .tensors import tensor_proto_to_tensor
+    return {k: tensor_proto_to_tensor(v)
+            for k, v in response.outputs.items()}

@@ -261,3 +261,14 @@ def test_metadata_signature_unpack():
             assert sig.outputs["logits"].name == "logits:0"
         finally:
             c.close()
+
+
+def test_decode_predict_response_helper(client):
+    import torch
+    from min_tfs_client_amd.client import decode_predict_response
+    x = np.arange(4, dtype=np.float32)
+    resp = client.predict_request("default", {"x": x})
+    outs = decode_predict_response(resp)
+    np.testing.assert_array_equal(outs["x"], x)
+    touts = decode_predict_response(resp, as_numpy=False)
+    assert torch.equal(touts["x"], torch.arange(4, dtype=torch.float32))
