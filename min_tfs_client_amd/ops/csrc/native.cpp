@@ -277,6 +277,55 @@ py::bytes echo_predict(py::buffer data) {
 
 }  // namespace
 
+// echo directly between two buffers (shm identity fast path, the
+// counterpart of echo_predict for the gRPC raw handler): parses the
+// request from src, writes the response into dst, memcpying payloads
+// once. Returns response length.
+uint64_t echo_predict_into(py::buffer src, uint64_t src_len,
+                           py::buffer dst) {
+  py::buffer_info si = src.request();
+  py::buffer_info di = dst.request(true);
+  TORCH_CHECK(uint64_t(si.size) >= src_len, "src too short");
+  auto parsed = tfswire::parse_predict_message(
+      static_cast<const uint8_t*>(si.ptr), size_t(src_len), true);
+  std::vector<std::string> names;
+  std::vector<tfswire::TensorMeta> metas;
+  std::vector<const uint8_t*> payloads;
+  for (auto& t : parsed.tensors) {
+    TORCH_CHECK(t.content != nullptr,
+                "echo_predict_into requires tensor_content inputs");
+    std::string name = t.name;
+    const std::string suffix = "_input";
+    if (name.size() > suffix.size() &&
+        name.compare(name.size() - suffix.size(), suffix.size(), suffix)
+            == 0) {
+      name = name.substr(0, name.size() - suffix.size()) + "_output";
+    }
+    names.push_back(std::move(name));
+    metas.push_back({t.dtype, t.shape, t.content_bytes});
+    payloads.push_back(t.content);
+  }
+  const std::string sig = parsed.model_spec.signature_name.empty()
+      ? "serving_default" : parsed.model_spec.signature_name;
+  auto plan = tfswire::plan_predict_message(
+      false, parsed.model_spec.name, parsed.model_spec.version, sig,
+      names, metas);
+  TORCH_CHECK(plan.total_size <= uint64_t(di.size),
+              "dst slot too small for echo response");
+  auto* buf = static_cast<uint8_t*>(di.ptr);
+  tfswire::write_predict_message(buf, plan, false, parsed.model_spec.name,
+                                 parsed.model_spec.version, sig, names,
+                                 metas);
+  {
+    py::gil_scoped_release release;
+    for (size_t i = 0; i < payloads.size(); ++i) {
+      std::memcpy(buf + plan.spans[i].offset, payloads[i],
+                  plan.spans[i].nbytes);
+    }
+  }
+  return plan.total_size;
+}
+
 // serialize into a caller-provided buffer (shared-memory transport):
 // returns bytes written; throws if capacity is too small.
 uint64_t serialize_predict_into(py::buffer dst, bool is_request,
@@ -388,7 +437,9 @@ py::tuple parse_predict_spans(py::buffer data, bool is_request) {
   spec["version"] = parsed.model_spec.version;
   spec["signature_name"] = parsed.model_spec.signature_name;
   spec["version_label"] = parsed.model_spec.version_label;
-  return py::make_tuple(spec, spans);
+  py::list filt;
+  for (auto& f : parsed.output_filter) filt.append(py::str(f));
+  return py::make_tuple(spec, spans, filt);
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -426,6 +477,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("echo_predict", &echo_predict, py::arg("data"));
   m.def("parse_predict_spans", &parse_predict_spans, py::arg("data"),
         py::arg("is_request") = false);
+  m.def("echo_predict_into", &echo_predict_into, py::arg("src"),
+        py::arg("src_len"), py::arg("dst"));
   m.def("serialize_predict_into", &serialize_predict_into,
         py::arg("dst"), py::arg("is_request"), py::arg("model_name"),
         py::arg("version"), py::arg("signature"), py::arg("names"),

@@ -267,11 +267,24 @@ class ShmListener:
         self._native.shm_wait_value(resp.buf, _STATE_OFF, _IDLE, 10.0)
         rlen = _read_len(req.buf)
         try:
-            spec, inputs, _filter = self._native.parse_predict_request(
-                req.buf[_HEADER:_HEADER + rlen], self.device, 1)
+            # cheap span parse first: spec + filter without tensor copies
+            spec, _spans, _filter = self._native.parse_predict_spans(
+                req.buf[_HEADER:_HEADER + rlen], True)
             version = spec["version"] if spec["version"] >= 0 else None
             label = spec.get("version_label") or None
             servable = self.manager.get(spec["name"], version, label)
+            if getattr(servable, "is_identity", False) and not _filter:
+                # identity fast path (mirror of the raw gRPC handler's
+                # echo_predict): payloads memcpy req->resp once
+                n = self._native.echo_predict_into(
+                    req.buf[_HEADER:_HEADER + rlen], rlen,
+                    resp.buf[_HEADER:])
+                _write_len(resp.buf, n)
+                self._native.shm_store_value(req.buf, _STATE_OFF, _IDLE)
+                self._native.shm_store_value(resp.buf, _STATE_OFF, _READY)
+                return
+            _spec2, inputs, _filter = self._native.parse_predict_request(
+                req.buf[_HEADER:_HEADER + rlen], self.device, 1)
             outputs = servable(inputs)
             if _filter:
                 outputs = {k: v for k, v in outputs.items()

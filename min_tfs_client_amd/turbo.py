@@ -155,7 +155,7 @@ class TurboPredictClient:
                 enum: TF_TO_TORCH_MAPPING[name]
                 for enum, name in ENUM_TO_TF_MAPPING.items()
                 if name in TF_TO_TORCH_MAPPING}
-        _spec, spans = self._native.parse_predict_spans(resp, False)
+        _spec, spans, _filt = self._native.parse_predict_spans(resp, False)
         out = {}
         for d in spans:
             if d is None:
