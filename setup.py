@@ -44,6 +44,7 @@ setup(
         include=["min_tfs_client_amd", "min_tfs_client_amd.*",
                  "min_tfs_client", "tensorflow", "tensorflow.*",
                  "tensorflow_serving", "tensorflow_serving.*"]),
+    package_data={"min_tfs_client_amd": ["py.typed"]},
     ext_modules=[ext],
     install_requires=["numpy", "grpcio>=1.21", "protobuf>=3.8", "torch"],
     python_requires=">=3.10",

@@ -84,3 +84,62 @@ def test_dp_predict_two_ranks():
         p.join(timeout=30)
     for rank, status in results:
         assert status == "ok", f"rank {rank}: {status}"
+
+
+def _dp_shm_worker(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        import torch.distributed as dist
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        import sys
+        root = os.path.dirname(os.path.dirname(
+            os.path.dirname(os.path.abspath(__file__))))
+        if root not in sys.path:
+            sys.path.insert(0, root)
+        from min_tfs_client_amd.parallel import DataParallelPredictor
+        from min_tfs_client_amd.server import ModelManager, identity_servable
+        from min_tfs_client_amd.shm import ShmListener, ShmPredictClient
+
+        mgr = ModelManager()
+        mgr.load("m", identity_servable(), version=1)
+        hs = f"/tmp/dp_shm_{os.getpid()}_{rank}"
+        with ShmListener(mgr, hs):
+            with ShmPredictClient(hs, slot_bytes=8 << 20) as client:
+                dp = DataParallelPredictor(client, device="cpu")
+                full = None
+                if rank == 0:
+                    torch.manual_seed(0)
+                    full = {"x": torch.randn(9, 4)}
+                out = dp.predict("m", full)
+                assert out["x"].shape == (9, 4)
+                if rank == 0:
+                    assert torch.equal(out["x"], full["x"])
+        dist.destroy_process_group()
+        q.put((rank, "ok"))
+    except Exception as e:  # noqa: BLE001
+        import traceback
+        q.put((rank, f"FAIL: {e}\n{traceback.format_exc()}"))
+
+
+@pytest.mark.timeout(180)
+def test_dp_predict_over_shm_transport():
+    """Config-4 composition with the shm local transport — the optimal
+    multi-GPU deployment shape (RCCL scatter/all-gather between ranks,
+    shared-memory hop to each rank's local server)."""
+    import socket
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    ctx = multiprocessing.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_dp_shm_worker, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=150) for _ in range(2)]
+    for p in procs:
+        p.join(timeout=30)
+    for rank, status in results:
+        assert status == "ok", f"rank {rank}: {status}"
