@@ -251,3 +251,26 @@ def test_tracer_disabled_is_noop():
     with trace_span("nothing"):
         pass
     assert t.export("/dev/null") == 0
+
+
+# -- shm native helpers ------------------------------------------------------
+
+def test_shm_wait_and_store_helpers():
+    import pytest as _pytest
+    native = _pytest.importorskip("min_tfs_client_amd._native")
+    import threading
+    buf = bytearray(64)
+    mv = memoryview(buf)
+    # timeout returns sentinel
+    assert native.shm_wait_value(mv, 0, 7, 0.05) == 0xFFFFFFFF
+    # store+wait round trip across a thread
+    def setter():
+        import time
+        time.sleep(0.05)
+        native.shm_store_value(mv, 0, 7)
+    t = threading.Thread(target=setter)
+    t.start()
+    assert native.shm_wait_value(mv, 0, 7, 5.0) == 7
+    t.join()
+    # immediate hit
+    assert native.shm_wait_value(mv, 0, 7, 0.01) == 7
