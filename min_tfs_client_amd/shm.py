@@ -25,12 +25,15 @@ a unix socket is vs TCP).
 from __future__ import annotations
 
 import json
+import logging
 import mmap
 import os
 import threading
 import time
 import uuid
 from typing import Dict, Optional
+
+logger = logging.getLogger("mi355x_tfs.shm")
 
 try:
     import torch
@@ -233,6 +236,9 @@ class ShmListener:
                     os.unlink(path)
                 except (OSError, json.JSONDecodeError):
                     continue
+                except Exception:  # noqa: BLE001 - accept loop must survive
+                    logger.exception("shm handshake %s failed", entry)
+                    continue
                 t = threading.Thread(target=self._serve_conn,
                                      args=(hello,), daemon=True)
                 t.start()
@@ -242,7 +248,8 @@ class ShmListener:
         try:
             req = _Segment(hello["req"], 0, create=False)
             resp = _Segment(hello["resp"], 0, create=False)
-        except FileNotFoundError:
+        except (FileNotFoundError, KeyError, ValueError):
+            logger.warning("shm connection %s not attachable", hello)
             return
         try:
             _write_len(resp.buf, 0xA110)  # attach ack
