@@ -175,3 +175,36 @@ def test_versioned_model_server_config():
     # Specific is field 102 inside ServableVersionPolicy
     assert list(back.model_config_list.config[0]
                 .model_version_policy.specific.versions) == [1, 3]
+
+
+def test_schema_builder_helpers():
+    """wire/schema.py builder primitives produce valid descriptors."""
+    from min_tfs_client_amd.wire.schema import (
+        enum, field, map_field, message, proto_file, service)
+    e = enum("E", [("A", 0), ("B", 5)])
+    assert [v.number for v in e.value] == [0, 5]
+    entry, f = map_field("m", 3, "string", "int64", ".pkg.Msg")
+    assert entry.options.map_entry is True
+    assert f.type_name == ".pkg.Msg.MEntry"
+    m = message("Msg", fields=[field("x", 1, "int32")], nested=[entry],
+                oneofs=["choice"])
+    assert m.oneof_decl[0].name == "choice"
+    svc = service("S", [("Do", ".pkg.Req", ".pkg.Resp")])
+    assert svc.method[0].input_type == ".pkg.Req"
+    fdp = proto_file("pkg/test_probe_unused.proto", "pkg", messages=[m],
+                     enums=[e], services=[svc])
+    assert fdp.syntax == "proto3"
+    assert fdp.message_type[0].name == "Msg"
+
+
+def test_all_wire_classes_instantiable():
+    """Every exported message class constructs and serializes empty."""
+    from min_tfs_client_amd.wire import messages as m
+    count = 0
+    for name in dir(m):
+        cls = getattr(m, name)
+        if isinstance(cls, type) and hasattr(cls, "SerializeToString") \
+                and hasattr(cls, "DESCRIPTOR"):
+            cls().SerializeToString()
+            count += 1
+    assert count >= 40
