@@ -126,3 +126,15 @@ def test_shm_segments_cleaned_up(listener):
     c.close()
     assert not os.path.exists(req_path)
     assert not os.path.exists(resp_path)
+
+
+def test_shm_connect_timeout_no_server(tmp_path):
+    import time
+    t0 = time.monotonic()
+    with pytest.raises(TimeoutError, match="did not attach"):
+        ShmPredictClient(str(tmp_path / "nobody"), slot_bytes=1 << 20,
+                         connect_timeout=0.5)
+    assert time.monotonic() - t0 < 5
+    # segments were unlinked on the failed connect
+    import glob
+    assert not glob.glob("/dev/shm/mi355x_req_*")
