@@ -108,6 +108,9 @@ class ShmPredictClient:
         self._native = require_native()
         self.slot_bytes = slot_bytes
         conn = uuid.uuid4().hex[:12]
+        # _closed must exist before anything can fail: close() after a
+        # failed connect must still unlink the segments
+        self._closed = False
         self._req = _Segment(f"/dev/shm/mi355x_req_{conn}",
                              _HEADER + slot_bytes, create=True)
         self._resp = _Segment(f"/dev/shm/mi355x_resp_{conn}",
@@ -130,7 +133,6 @@ class ShmPredictClient:
                 raise TimeoutError("shm server did not attach")
             time.sleep(0.005)
         _write_len(self._resp.buf, 0)
-        self._closed = False
 
     def predict(self, model_name: str, inputs: Dict[str, "torch.Tensor"],
                 timeout: float = 60.0,
@@ -164,7 +166,7 @@ class ShmPredictClient:
             self._native.shm_store_value(self._resp.buf, _STATE_OFF, _IDLE)
 
     def close(self, unlink: bool = True):
-        if getattr(self, "_closed", True):
+        if getattr(self, "_closed", True) or not hasattr(self, "_req"):
             return
         self._closed = True
         try:

@@ -128,13 +128,25 @@ def test_shm_segments_cleaned_up(listener):
     assert not os.path.exists(resp_path)
 
 
-def test_shm_connect_timeout_no_server(tmp_path):
+def test_shm_connect_timeout_no_server(tmp_path, monkeypatch):
     import time
+    created = []
+    from min_tfs_client_amd import shm as shm_mod
+    orig = shm_mod._Segment
+
+    class Tracking(orig):
+        def __init__(self, path, size, create):
+            super().__init__(path, size, create)
+            if create:
+                created.append(path)
+
+    monkeypatch.setattr(shm_mod, "_Segment", Tracking)
     t0 = time.monotonic()
     with pytest.raises(TimeoutError, match="did not attach"):
         ShmPredictClient(str(tmp_path / "nobody"), slot_bytes=1 << 20,
                          connect_timeout=0.5)
     assert time.monotonic() - t0 < 5
-    # segments were unlinked on the failed connect
-    import glob
-    assert not glob.glob("/dev/shm/mi355x_req_*")
+    # exactly this client's segments were unlinked on the failed connect
+    assert len(created) == 2
+    for path in created:
+        assert not os.path.exists(path)
