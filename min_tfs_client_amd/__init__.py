@@ -41,4 +41,7 @@ def __getattr__(name):
     if name == "DataParallelPredictor":
         from .parallel import DataParallelPredictor
         return DataParallelPredictor
+    if name == "ShmPredictClient":
+        from .shm import ShmPredictClient
+        return ShmPredictClient
     raise AttributeError(name)

@@ -140,6 +140,8 @@ class ShmPredictClient:
                 signature_name: str = "",
                 output_device: Optional[str] = None,
                 copy_mode: int = 1) -> Dict[str, "torch.Tensor"]:
+        if self._closed:
+            raise RuntimeError("ShmPredictClient is closed")
         names = list(inputs.keys())
         tensors = [inputs[k] for k in names]
         n = self._native.serialize_predict_into(

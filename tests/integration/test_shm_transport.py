@@ -150,3 +150,10 @@ def test_shm_connect_timeout_no_server(tmp_path, monkeypatch):
     assert len(created) == 2
     for path in created:
         assert not os.path.exists(path)
+
+
+def test_predict_after_close_raises(listener):
+    c = ShmPredictClient(listener.dir, slot_bytes=1 << 20)
+    c.close()
+    with pytest.raises(RuntimeError, match="closed"):
+        c.predict("m", {"x": torch.ones(1)})
