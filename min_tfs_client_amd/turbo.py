@@ -214,10 +214,15 @@ class TurboPredictClient:
             off += n
         dev = str(output_device) if output_device is not None else "cpu"
         parts = []
-        for fut in futs:
-            _s, outputs, _ = self._native.parse_predict_response(
-                fut.result(), dev, copy_mode)
-            parts.append(outputs)
+        try:
+            for fut in futs:
+                _s, outputs, _ = self._native.parse_predict_response(
+                    fut.result(), dev, copy_mode)
+                parts.append(outputs)
+        except Exception:
+            for fut in futs:
+                fut.cancel()
+            raise
         merged = {}
         for k in parts[0]:
             vals = [p[k] for p in parts]

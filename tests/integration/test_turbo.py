@@ -244,3 +244,16 @@ def test_multi_target_client():
             n1 = s1.metrics.latency_quantiles("predict").get("count", 0)
             n2 = s2.metrics.latency_quantiles("predict").get("count", 0)
             assert n1 > 0 and n2 > 0
+
+
+def test_predict_sharded_error_cancels(raw_server):
+    """A failing shard rpc propagates and cancels the siblings."""
+    import grpc
+    with TurboPredictClient(raw_server.address, num_channels=2) as client:
+        with pytest.raises(grpc.RpcError):
+            client.predict_sharded("missing_model",
+                                   {"x": torch.randn(8, 2)}, shards=2)
+        # client still usable
+        out = client.predict_sharded("default",
+                                     {"x": torch.ones(4, 2)}, shards=2)
+        assert torch.equal(out["x"], torch.ones(4, 2))
