@@ -70,8 +70,9 @@ def _masked_crc(data: bytes) -> int:
     return ((crc >> 15) | (crc << 17)) + 0xA282EAD8 & 0xFFFFFFFF
 
 
-def write_tfrecord(path: str, records: List[bytes]) -> None:
-    with open(path, "wb") as f:
+def write_tfrecord(path: str, records: List[bytes],
+                   append: bool = False) -> None:
+    with open(path, "ab" if append else "wb") as f:
         for rec in records:
             length = struct.pack("<Q", len(rec))
             f.write(length)

@@ -26,15 +26,27 @@ class FileLogCollector:
         self._lock = threading.Lock()
         self._buffers: Dict[str, list] = {}
 
-    def collect(self, model_name: str, record: bytes):
+    def collect(self, model_name: str, record: bytes,
+                flush_every: int = 256):
         with self._lock:
-            self._buffers.setdefault(model_name, []).append(record)
+            buf = self._buffers.setdefault(model_name, [])
+            buf.append(record)
+            if len(buf) >= flush_every:
+                self._flush_locked(model)
+
+    def _flush_locked(self, model: str):
+        records = self._buffers.get(model) or []
+        if records:
+            # append: a long-running server must not re-buffer or rewrite
+            # its whole log history
+            write_tfrecord(f"{self.filename_prefix}.{model}.log", records,
+                           append=True)
+            self._buffers[model] = []
 
     def flush(self):
         with self._lock:
-            for model, records in self._buffers.items():
-                write_tfrecord(f"{self.filename_prefix}.{model}.log",
-                               records)
+            for model in list(self._buffers):
+                self._flush_locked(model)
 
 
 class RequestLogger:

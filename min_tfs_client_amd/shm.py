@@ -247,6 +247,8 @@ class ShmListener:
                                      args=(hello,), daemon=True)
                 t.start()
                 self._threads.append(t)
+            # prune finished connection threads (long-running servers)
+            self._threads = [t for t in self._threads if t.is_alive()]
 
     def _serve_conn(self, hello):
         try:

@@ -274,3 +274,29 @@ def test_shm_wait_and_store_helpers():
     t.join()
     # immediate hit
     assert native.shm_wait_value(mv, 0, 7, 0.01) == 7
+
+
+def test_log_collector_appends_across_flushes(tmp_path):
+    collector = FileLogCollector(str(tmp_path / "log"))
+    logger = RequestLogger(collector, sampling_rate=1.0)
+    req, resp = _predict_pair()
+    logger.log_predict(req, resp)
+    collector.flush()
+    logger.log_predict(req, resp)
+    collector.flush()
+    collector.flush()  # idempotent on empty buffer
+    records = read_tfrecord(str(tmp_path / "log.m.log"))
+    assert len(records) == 2
+
+
+def test_log_collector_auto_flush(tmp_path):
+    collector = FileLogCollector(str(tmp_path / "auto"))
+    logger = RequestLogger(collector, sampling_rate=1.0)
+    req, resp = _predict_pair()
+    for _ in range(300):
+        logger.log_predict(req, resp)
+    # 256-record auto-flush happened without an explicit flush()
+    records = read_tfrecord(str(tmp_path / "auto.m.log"))
+    assert len(records) >= 256
+    collector.flush()
+    assert len(read_tfrecord(str(tmp_path / "auto.m.log"))) == 300
