@@ -32,7 +32,7 @@ class FileLogCollector:
             buf = self._buffers.setdefault(model_name, [])
             buf.append(record)
             if len(buf) >= flush_every:
-                self._flush_locked(model)
+                self._flush_locked(model_name)
 
     def _flush_locked(self, model: str):
         records = self._buffers.get(model) or []
