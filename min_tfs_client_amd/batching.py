@@ -104,6 +104,8 @@ class BatchingServable(Servable):
                 f"{self.max_batch_size}")
         p = _Pending(inputs, batch)
         with self._lock:
+            if self._closed:
+                raise RuntimeError("batching servable is shut down")
             if len(self._queue) >= self._max_queue:
                 raise RuntimeError("batching queue full")
             self._queue.append(p)
@@ -147,6 +149,13 @@ class BatchingServable(Servable):
         while True:
             taken = self._take_batch()
             if not taken:
+                # drain: fail anything that raced past the closed check
+                with self._lock:
+                    leftover = self._queue
+                    self._queue = []
+                for p in leftover:
+                    p.error = RuntimeError("batching servable is shut down")
+                    p.event.set()
                 return
             try:
                 self._run_batch(taken)

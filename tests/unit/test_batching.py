@@ -151,3 +151,12 @@ def test_unload_closes_batcher_thread():
     mgr.unload("m", 1)
     b._thread.join(timeout=5)
     assert not b._thread.is_alive()
+
+
+def test_closed_batcher_rejects_and_drains():
+    inner = CountingInner()
+    b = BatchingServable(inner, max_batch_size=4, batch_timeout_s=0.01)
+    b.close()
+    b._thread.join(timeout=5)
+    with pytest.raises(RuntimeError, match="shut down"):
+        b({"x": np.ones((1, 1), dtype=np.float32)})
