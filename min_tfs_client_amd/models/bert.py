@@ -65,6 +65,10 @@ class BertEncoder(nn.Module):
 
     def forward(self, input_ids, attention_mask=None):
         B, S = input_ids.shape
+        if S > self.cfg.max_pos:
+            raise ValueError(
+                f"sequence length {S} exceeds max_position_embeddings "
+                f"{self.cfg.max_pos}")
         pos_ids = torch.arange(S, device=input_ids.device)
         x = self.tok(input_ids.long()) + self.pos(pos_ids) \
             + self.typ(torch.zeros_like(input_ids.long()))

@@ -1,5 +1,6 @@
 """Model-family sanity on CPU (tiny shapes; full shapes are GPU tests)."""
 import numpy as np
+import pytest
 import torch
 
 from min_tfs_client_amd.models import (
@@ -53,3 +54,23 @@ def test_bert_attention_mask_effect():
     # masking must change the pooled output
     assert not np.allclose(full["pooled_output"].numpy(),
                            half["pooled_output"].numpy())
+
+
+def test_bert_seq_too_long_raises():
+    s = bert_servable()
+    ids = np.zeros((1, 513), dtype=np.int32)
+    with pytest.raises(ValueError, match="sequence length"):
+        s({"input_ids": ids, "attention_mask": np.ones((1, 513), np.int32)})
+
+
+def test_bert_without_mask():
+    s = bert_servable()
+    ids = np.random.randint(0, 30522, (1, 8), dtype=np.int32)
+    out = s({"input_ids": ids})
+    assert out["pooled_output"].shape == (1, 768)
+
+
+def test_resnet_wrong_channels_raises():
+    s = resnet50_servable()
+    with pytest.raises(Exception):
+        s({"images": np.zeros((1, 4, 32, 32), np.float32)})
