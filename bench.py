@@ -73,7 +73,7 @@ def make_inputs(cfg: str, device, batch_override=None):
         return {"input_ids": ids, "attention_mask": mask}
     if cfg == "scalar":
         return {"x": torch.zeros((), dtype=torch.float32, device=device)}
-    if cfg == "bf16pack":
+    if cfg in ("bf16pack", "fused"):
         b = batch_override or 32
         return {"images": torch.randn(b, 3, 224, 224, device=device,
                                       dtype=torch.bfloat16)}
@@ -87,7 +87,11 @@ def main():
     ap.add_argument("--steps", type=int, default=64)
     ap.add_argument("--warmup", type=int, default=16)
     ap.add_argument("--bench-config", default="resnet50",
-                    choices=["resnet50", "bert", "scalar", "bf16pack"])
+                    choices=["resnet50", "bert", "scalar", "bf16pack",
+                             "fused"],
+                    help="fused = BASELINE config 5: bf16 NCHW input, "
+                         "fused cast+NCHW->NHWC CDNA4 kernel before "
+                         "serialize, fp32 NHWC on the wire")
     ap.add_argument("--encoding", default="turbo",
                     choices=["turbo", "proto"],
                     help="turbo = C++ codec raw-bytes path; proto = "
@@ -240,15 +244,19 @@ def main():
             num_channels=max(min(args.pipeline, 8), args.shards,
                              args.servers))
 
+        transform = ({"images": ("nhwc", torch.float32)}
+                     if args.bench_config == "fused" else None)
+
         def step_fn(step_inputs):
             out_dev = f"cuda:{local_rank}" if has_gpu else "cpu"
-            if args.shards > 1:
+            if args.shards > 1 and transform is None:
                 return client.predict_sharded(
                     "default", step_inputs, shards=args.shards,
                     output_device=out_dev, copy_mode=args.copy_mode)
             return client.predict("default", step_inputs,
                                   output_device=out_dev,
-                                  copy_mode=args.copy_mode)
+                                  copy_mode=args.copy_mode,
+                                  transform=transform)
     else:
         from min_tfs_client_amd.client import TensorServingClient
         if args.transport == "unix":
@@ -385,7 +393,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": ("bf16" if args.bench_config == "bf16pack"
+            "dtype": ("bf16" if args.bench_config in ("bf16pack", "fused")
                       else "int32" if args.bench_config == "bert"
                       else "fp32"),
             "data": "synthetic",

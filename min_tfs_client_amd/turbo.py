@@ -119,12 +119,39 @@ class TurboPredictClient:
                 signature_name: str = "",
                 output_device: Optional[Union[str, "torch.device"]] = None,
                 copy_mode: int = 1,
-                zero_copy: bool = False) -> Dict[str, "torch.Tensor"]:
+                zero_copy: bool = False,
+                transform: Optional[Dict[str, tuple]] = None
+                ) -> Dict[str, "torch.Tensor"]:
         """One Predict round trip. ``output_device``: where response
         tensors land ("cpu" default; "cuda:N" unpacks over the staging
         pipeline straight to HBM). ``zero_copy=True`` (CPU outputs only)
         returns read-only tensor views borrowing the response buffer —
-        no copy at all."""
+        no copy at all.
+
+        ``transform``: per-input pre-pack device transform, fused into ONE
+        CDNA4 kernel on the GPU (BASELINE config 5): {name: (layout,
+        dtype)} with layout in {"nhwc", "nchw", None}. E.g.
+        ``transform={"images": ("nhwc", torch.float32)}`` converts a bf16
+        NCHW batch to fp32 NHWC in a single LDS-tiled kernel before the
+        bytes leave HBM."""
+        if transform:
+            inputs = dict(inputs)
+            for name, (layout, dtype) in transform.items():
+                t = inputs[name]
+                if t.is_cuda:
+                    from . import ops
+                    if layout == "nhwc":
+                        inputs[name] = ops.nchw_to_nhwc(t, dtype or t.dtype)
+                    elif layout == "nchw":
+                        inputs[name] = ops.nhwc_to_nchw(t, dtype or t.dtype)
+                    elif dtype is not None and dtype != t.dtype:
+                        inputs[name] = ops.cast(t, dtype)
+                else:  # CPU fallback: plain torch (no HIP available)
+                    if layout == "nhwc":
+                        t = t.permute(0, 2, 3, 1).contiguous()
+                    elif layout == "nchw":
+                        t = t.permute(0, 3, 1, 2).contiguous()
+                    inputs[name] = t.to(dtype) if dtype else t
         with trace_span("turbo.serialize", model=model_name,
                         bytes=sum(t.numel() * t.element_size()
                                   for t in inputs.values())):

@@ -65,3 +65,14 @@ def test_gpu_client_python_proto_path(server):
     assert len(proto.tensor_content) == x.numel() * 4
     import numpy as np
     assert proto.tensor_content == x.cpu().numpy().tobytes()
+
+
+def test_predict_transform_fused_kernel(server):
+    """BASELINE config 5 as a one-liner: the transform= arg runs the fused
+    bf16 NCHW -> fp32 NHWC CDNA4 kernel before serialize."""
+    with TurboPredictClient(server.address) as client:
+        x = torch.randn(8, 3, 64, 64, device=DEV, dtype=torch.bfloat16)
+        out = client.predict("default", {"images": x}, output_device=DEV,
+                             transform={"images": ("nhwc", torch.float32)})
+        ref = x.permute(0, 2, 3, 1).contiguous().to(torch.float32)
+        assert torch.equal(out["images"], ref)

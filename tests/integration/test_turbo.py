@@ -257,3 +257,15 @@ def test_predict_sharded_error_cancels(raw_server):
         out = client.predict_sharded("default",
                                      {"x": torch.ones(4, 2)}, shards=2)
         assert torch.equal(out["x"], torch.ones(4, 2))
+
+
+def test_predict_transform_cpu_fallback(raw_server):
+    """transform= applies layout/dtype conversion before packing (CPU
+    fallback path; the GPU path runs the fused CDNA4 kernel)."""
+    with TurboPredictClient(raw_server.address) as c:
+        x = torch.randn(2, 3, 4, 5, dtype=torch.bfloat16)
+        out = c.predict("default", {"images": x},
+                        transform={"images": ("nhwc", torch.float32)})
+        ref = x.permute(0, 2, 3, 1).contiguous().to(torch.float32)
+        assert out["images"].shape == (2, 4, 5, 3)
+        assert torch.equal(out["images"], ref)
