@@ -94,18 +94,33 @@ class ServerRequestLogger:
 
     def __init__(self):
         self._loggers: Dict[str, RequestLogger] = {}
+        # observers called on every (re)configure — the native transport
+        # drops logged models from its C++ echo fast path so every logged
+        # request passes through the python handler
+        self._subscribers = []
+
+    def subscribe(self, callback) -> None:
+        self._subscribers.append(callback)
+
+    def logged_models(self):
+        return set(self._loggers)
 
     def configure(self, model_name: str, logging_config,
                   collector: Optional[FileLogCollector] = None):
         """logging_config: pb.LoggingConfig (or None to remove)."""
         if logging_config is None:
             self._loggers.pop(model_name, None)
-            return
-        prefix = (logging_config.log_collector_config.filename_prefix
-                  or f"/tmp/prediction_log_{model_name}")
-        rate = logging_config.sampling_config.sampling_rate or 1.0
-        self._loggers[model_name] = RequestLogger(
-            collector or FileLogCollector(prefix), sampling_rate=rate)
+        else:
+            prefix = (logging_config.log_collector_config.filename_prefix
+                      or f"/tmp/prediction_log_{model_name}")
+            rate = logging_config.sampling_config.sampling_rate or 1.0
+            self._loggers[model_name] = RequestLogger(
+                collector or FileLogCollector(prefix), sampling_rate=rate)
+        for cb in list(self._subscribers):
+            try:
+                cb()
+            except Exception:  # noqa: BLE001
+                pass
 
     def get(self, model_name: str) -> Optional[RequestLogger]:
         return self._loggers.get(model_name)

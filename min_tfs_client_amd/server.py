@@ -237,6 +237,75 @@ class ModelManager:
         with self._lock:
             return list(self._models)
 
+    def iter_available(self):
+        """Snapshot of (name, version, servable) for every AVAILABLE
+        version (used by the native transport's echo-table refresh)."""
+        out = []
+        with self._lock:
+            for name, versions in self._models.items():
+                for ver, v in versions.items():
+                    if v.state == STATE_AVAILABLE and v.servable is not None:
+                        out.append((name, ver, v.servable))
+        return out
+
+
+# ---------------------------------------------------------------------------
+# SignatureDef-driven request validation (predict_util.cc:66-146 semantics)
+# ---------------------------------------------------------------------------
+
+def validate_inputs_against_signature(servable, inputs) -> Optional[str]:
+    """Returns a TF-Serving-shaped INVALID_ARGUMENT message when the
+    request's input aliases do not match the servable's declared input
+    signature, else None. Servables without a declared input signature
+    accept any aliases (mirrors serving a SignatureDef-less model).
+
+    Message shapes match reference predict_util.cc:66-111
+    (VerifyRequestInputsSize / PreProcessPrediction).
+    """
+    sig_inputs = (servable.signature or {}).get("inputs")
+    if not sig_inputs:
+        return None
+    req = set(inputs)
+    sig = set(sig_inputs)
+    if len(req) != len(sig):
+        extra = sorted(req - sig)
+        missing = sorted(sig - req)
+        return (
+            f"input size does not match signature: {len(req)}!={len(sig)} "
+            f"len({{{','.join(sorted(req))}}}) != "
+            f"len({{{','.join(sorted(sig))}}}). "
+            f"Sent extra: {{{','.join(extra)}}}. "
+            f"Missing but required: {{{','.join(missing)}}}.")
+    for alias in sorted(req):
+        if alias not in sig:
+            return (
+                f"input tensor alias not found in signature: {alias}. "
+                f"Inputs expected to be in the set "
+                f"{{{','.join(sorted(sig))}}}.")
+    return None
+
+
+def validate_output_filter(servable, outputs, output_filter) -> Optional[str]:
+    """Validates Predict.output_filter entries against the declared output
+    signature (when present) or the produced outputs; unknown aliases and
+    duplicates are INVALID_ARGUMENT per predict_util.cc:119-136 (the
+    reference never silently drops a filter entry)."""
+    if not output_filter:
+        return None
+    sig_outputs = (servable.signature or {}).get("outputs")
+    known = set(sig_outputs) if sig_outputs else set(outputs)
+    seen = set()
+    for alias in output_filter:
+        if alias not in known:
+            return (
+                f"output tensor alias not found in signature: {alias} "
+                f"Outputs expected to be in the set "
+                f"{{{','.join(sorted(known))}}}.")
+        if alias in seen:
+            return f"duplicate output tensor alias: {alias}"
+        seen.add(alias)
+    return None
+
 
 # ---------------------------------------------------------------------------
 # Service implementations
@@ -303,10 +372,21 @@ class PredictionServiceImpl(PredictionServiceServicer):
         except Exception as e:
             _abort(context, grpc.StatusCode.INVALID_ARGUMENT,
                    f"tensor parsing error: {e}")
+        err = validate_inputs_against_signature(servable, inputs)
+        if err is not None:
+            _abort(context, grpc.StatusCode.INVALID_ARGUMENT, err)
         try:
             outputs = servable(inputs)
+        except ValueError as e:
+            # e.g. batch larger than max_batch_size (BatchingServable):
+            # TF-Serving surfaces these as INVALID_ARGUMENT, not INTERNAL
+            _abort(context, grpc.StatusCode.INVALID_ARGUMENT, str(e))
         except Exception as e:
             _abort(context, grpc.StatusCode.INTERNAL, str(e))
+        err = validate_output_filter(servable, outputs,
+                                     request.output_filter)
+        if err is not None:
+            _abort(context, grpc.StatusCode.INVALID_ARGUMENT, err)
         response = pb.PredictResponse()
         response.model_spec.CopyFrom(request.model_spec)
         response.model_spec.signature_name = (
@@ -507,10 +587,18 @@ def _raw_predict_handler(manager: ModelManager, device: str,
                 request_logger.log_predict(spec["name"], bytes(data),
                                            bytes(out))
             return out
+        err = validate_inputs_against_signature(servable, inputs)
+        if err is not None:
+            _abort(context, grpc.StatusCode.INVALID_ARGUMENT, err)
         try:
             outputs = servable(inputs)
+        except ValueError as e:
+            _abort(context, grpc.StatusCode.INVALID_ARGUMENT, str(e))
         except Exception as e:  # noqa: BLE001
             _abort(context, grpc.StatusCode.INTERNAL, str(e))
+        err = validate_output_filter(servable, outputs, _filter)
+        if err is not None:
+            _abort(context, grpc.StatusCode.INVALID_ARGUMENT, err)
         if _filter:
             outputs = {k: v for k, v in outputs.items() if k in _filter}
         names = list(outputs.keys())
@@ -554,13 +642,47 @@ class ModelServer:
                  raw_predict: bool = False,
                  device: str = "cpu",
                  address: Optional[str] = None,
-                 shm_handshake_dir: Optional[str] = None):
+                 shm_handshake_dir: Optional[str] = None,
+                 transport: str = "native"):
         from .utils.allocator import tune_malloc
         tune_malloc()
         from .request_logging import ServerRequestLogger
         self.manager = manager or ModelManager()
         self.metrics = MetricsRegistry()
         self.request_logger = ServerRequestLogger()
+        self.prediction_service = None
+        self.model_service = None
+        self._server = None
+        self._native = None
+        if transport == "native":
+            # C++ HTTP/2 gRPC server (native_transport.py): the Predict
+            # data plane stays off python-grpcio entirely
+            from .native_transport import NativeTransportServer
+            self.prediction_service = PredictionServiceImpl(
+                self.manager, output_encoding, self.metrics,
+                self.request_logger)
+            self.model_service = ModelServiceImpl(self.manager,
+                                                  servable_factory)
+            native_addr = address if address is not None \
+                else f"127.0.0.1:{port}"
+            self._native = NativeTransportServer(
+                self.manager, self.prediction_service, self.model_service,
+                native_addr, device=device, metrics=self.metrics,
+                request_logger=self.request_logger,
+                max_workers=max_workers,
+                output_encoding=output_encoding)
+            self.shm_listener = None
+            if shm_handshake_dir:
+                from .shm import ShmListener
+                self.shm_listener = ShmListener(self.manager,
+                                                shm_handshake_dir,
+                                                device=device)
+            self.address = native_addr
+            self.port = port
+            return
+        if transport != "grpcio":
+            raise ValueError(f"unknown transport {transport!r} "
+                             "(expected 'native' or 'grpcio')")
         self._server = grpc.server(
             futures.ThreadPoolExecutor(max_workers=max_workers),
             options=[("grpc.max_send_message_length", 1 << 30),
@@ -624,7 +746,12 @@ class ModelServer:
             self.address = f"127.0.0.1:{self.port}"
 
     def start(self) -> "ModelServer":
-        self._server.start()
+        if self._native is not None:
+            self.address = self._native.start()
+            if not self.address.startswith("unix:"):
+                self.port = int(self.address.rsplit(":", 1)[1])
+        else:
+            self._server.start()
         if self.shm_listener is not None:
             self.shm_listener.start()
         return self
@@ -632,7 +759,10 @@ class ModelServer:
     def stop(self, grace: Optional[float] = None) -> None:
         if self.shm_listener is not None:
             self.shm_listener.stop()
-        self._server.stop(grace)
+        if self._native is not None:
+            self._native.stop()
+        else:
+            self._server.stop(grace)
 
     def __enter__(self):
         return self.start()

@@ -45,6 +45,74 @@ def _identity(x: bytes) -> bytes:
     return x
 
 
+_INT_TO_STATUS = {code.value[0]: code for code in grpc.StatusCode}
+
+
+class NativeRpcError(grpc.RpcError):
+    """grpc.RpcError-compatible failure from the native transport (same
+    .code()/.details() surface as grpcio call errors)."""
+
+    def __init__(self, code_int: int, details: str):
+        super().__init__()
+        self._code = _INT_TO_STATUS.get(code_int, grpc.StatusCode.UNKNOWN)
+        self._details = details
+
+    def code(self):
+        return self._code
+
+    def details(self):
+        return self._details
+
+    def __str__(self):  # pragma: no cover - debugging aid
+        return (f"NativeRpcError(code={self._code}, "
+                f"details={self._details!r})")
+
+
+def _translate_native_error(fn):
+    def wrapped(*args, **kwargs):
+        from . import _transport
+        try:
+            return fn(*args, **kwargs)
+        except _transport.NativeRpcError as e:
+            raise NativeRpcError(e.code_int, e.details) from None
+
+    return wrapped
+
+
+class _NativeFuture:
+    """grpc.Future-shaped handle over a native in-flight call."""
+
+    def __init__(self, channel, call_id: int, timeout: float):
+        self._channel = channel
+        self._id = call_id
+        self._timeout = timeout
+
+    @_translate_native_error
+    def result(self):
+        return self._channel.wait(self._id, self._timeout)
+
+    def cancel(self):  # in-flight unary calls finish server-side anyway
+        return False
+
+
+class _NativeStub:
+    """grpcio multicallable-shaped adapter over a native GrpcChannel.
+    Responses are zero-copy OwnedBuf buffers (the C++ parse and
+    torch.frombuffer both accept them directly)."""
+
+    def __init__(self, channel, path: str):
+        self._channel = channel
+        self._path = path
+
+    @_translate_native_error
+    def __call__(self, blob, timeout=60.0):
+        return self._channel.call(self._path, blob, timeout or 0.0)
+
+    def future(self, blob, timeout=60.0):
+        call_id = self._channel.start(self._path, blob, timeout or 0.0)
+        return _NativeFuture(self._channel, call_id, timeout or 0.0)
+
+
 class TurboPredictClient:
     """Raw-bytes Predict client over the C++ codec.
 
@@ -59,28 +127,54 @@ class TurboPredictClient:
     def __init__(self, target,
                  credentials: Optional[grpc.ChannelCredentials] = None,
                  options: Optional[list] = None,
-                 num_channels: int = 1):
+                 num_channels: int = 1,
+                 backend: str = "auto"):
+        """``backend``: "native" = the C++ HTTP/2 transport (~2 copies per
+        hop, no python-grpcio on the data plane); "grpcio" = python gRPC
+        (required for TLS); "auto" = native when available and no
+        credentials were given."""
         self._native = require_native()
         tune_malloc()  # large wire buffers: arena reuse, no per-call mmap
-        opts = _CHANNEL_OPTS + (options or [])
+        if backend == "auto":
+            backend = "grpcio" if credentials is not None else "native"
+            if backend == "native":
+                try:
+                    from . import _transport  # noqa: F401
+                except Exception:
+                    backend = "grpcio"
+        self.backend = backend
         targets = [target] if isinstance(target, str) else list(target)
+        n = max(len(targets), max(1, num_channels))
         self._channels = []
         self._stubs = []
-        n = max(len(targets), max(1, num_channels))
-        for i in range(n):
-            # separate HTTP/2 connections (no shared subchannel) ->
-            # parallel transport for pipelined requests
-            copts = opts + [("grpc.use_local_subchannel_pool", 1),
-                            ("grpc.channel_id", i)]
-            tgt = targets[i % len(targets)]
-            if credentials:
-                ch = grpc.secure_channel(tgt, credentials, options=copts)
-            else:
-                ch = grpc.insecure_channel(tgt, options=copts)
-            self._channels.append(ch)
-            self._stubs.append(ch.unary_unary(
-                _PREDICT_PATH, request_serializer=_identity,
-                response_deserializer=_identity))
+        if backend == "native":
+            if credentials is not None:
+                raise ValueError("backend='native' is cleartext (h2c); use "
+                                 "backend='grpcio' for TLS")
+            from . import _transport
+            for i in range(n):
+                ch = _transport.GrpcChannel(targets[i % len(targets)])
+                self._channels.append(ch)
+                self._stubs.append(_NativeStub(ch, _PREDICT_PATH))
+        elif backend == "grpcio":
+            opts = _CHANNEL_OPTS + (options or [])
+            for i in range(n):
+                # separate HTTP/2 connections (no shared subchannel) ->
+                # parallel transport for pipelined requests
+                copts = opts + [("grpc.use_local_subchannel_pool", 1),
+                                ("grpc.channel_id", i)]
+                tgt = targets[i % len(targets)]
+                if credentials:
+                    ch = grpc.secure_channel(tgt, credentials,
+                                             options=copts)
+                else:
+                    ch = grpc.insecure_channel(tgt, options=copts)
+                self._channels.append(ch)
+                self._stubs.append(ch.unary_unary(
+                    _PREDICT_PATH, request_serializer=_identity,
+                    response_deserializer=_identity))
+        else:
+            raise ValueError(f"unknown backend {backend!r}")
         self._channel = self._channels[0]
         self._predict = self._stubs[0]
         self._rr = 0

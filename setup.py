@@ -13,7 +13,11 @@ from setuptools import find_namespace_packages, setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
-from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+from torch.utils.cpp_extension import (  # noqa: E402
+    BuildExtension,
+    CppExtension,
+    CUDAExtension,
+)
 
 _CSRC = os.path.join("min_tfs_client_amd", "ops", "csrc")
 
@@ -32,6 +36,13 @@ ext = CUDAExtension(
     },
 )
 
+# native gRPC (HTTP/2) transport: pure C++ (sockets + HPACK), no HIP
+transport_ext = CppExtension(
+    name="min_tfs_client_amd._transport",
+    sources=[os.path.join(_CSRC, "grpc_transport.cpp")],
+    extra_compile_args={"cxx": ["-O3", "-std=c++17"]},
+)
+
 setup(
     name="min-tfs-client-amd",
     version="0.1.0",
@@ -45,7 +56,7 @@ setup(
                  "min_tfs_client", "tensorflow", "tensorflow.*",
                  "tensorflow_serving", "tensorflow_serving.*"]),
     package_data={"min_tfs_client_amd": ["py.typed"]},
-    ext_modules=[ext],
+    ext_modules=[ext, transport_ext],
     install_requires=["numpy", "grpcio>=1.21", "protobuf>=3.8", "torch"],
     python_requires=">=3.10",
     cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},

@@ -108,7 +108,9 @@ def test_ssl_channel(tmp_path):
     key = open(os.path.join(certs, "server.key"), "rb").read()
     crt = open(os.path.join(certs, "server.crt"), "rb").read()
 
-    srv = ModelServer(port=0)
+    # TLS serving stays on the grpcio transport (the native
+    # C++ transport is cleartext h2c)
+    srv = ModelServer(port=0, transport="grpcio")
     srv.manager.load("default", identity_servable(), version=1)
     creds = grpc.ssl_server_credentials([(key, crt)])
     port = srv._server.add_secure_port("localhost:0", creds)
