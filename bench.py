@@ -96,6 +96,11 @@ def main():
                     choices=["turbo", "proto"],
                     help="turbo = C++ codec raw-bytes path; proto = "
                          "python-protobuf client (reference-style)")
+    ap.add_argument("--grpc-impl", default="native",
+                    choices=["native", "grpcio"],
+                    help="gRPC stack for the turbo path: native = the "
+                         "C++ HTTP/2 transport (~2 copies/hop); grpcio = "
+                         "python gRPC (round-1 baseline, A/B)")
     ap.add_argument("--copy-mode", type=int, default=1,
                     help="0 = pinned-staged pipelined copies, 1 = direct "
                          "pageable hipMemcpy (A/B)")
@@ -242,7 +247,8 @@ def main():
         client = TurboPredictClient(
             address,
             num_channels=max(min(args.pipeline, 8), args.shards,
-                             args.servers))
+                             args.servers),
+            backend=args.grpc_impl)
 
         transform = ({"images": ("nhwc", torch.float32)}
                      if args.bench_config == "fused" else None)
@@ -378,7 +384,9 @@ def main():
     # whole-job aggregate: one request per rank per step
     reqs_per_s = n_gpus * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1e3
+    lat.sort()
     p50_ms = statistics.median(lat) * 1e3
+    p99_ms = lat[min(len(lat) - 1, int(0.99 * len(lat)))] * 1e3
 
     if rank == 0:
         result = {
@@ -390,6 +398,7 @@ def main():
             "warmup": args.warmup,
             "ms_per_step": round(ms_per_step, 3),
             "p50_ms_rtt": round(p50_ms, 3),
+            "p99_ms_rtt": round(p99_ms, 3),
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
@@ -409,6 +418,7 @@ def main():
                 "pipeline": args.pipeline,
                 "shards": args.shards,
                 "encoding": args.encoding,
+                "grpc_impl": args.grpc_impl,
                 "copy_mode": args.copy_mode,
                 "transport": args.transport,
                 "servers_per_rank": args.servers,
