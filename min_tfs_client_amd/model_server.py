@@ -90,12 +90,30 @@ def make_server(args) -> tuple:
                 batch_timeout_s=args.batch_timeout_micros / 1e6,
                 allowed_batch_sizes=allowed)
 
+    from .server import ModelManager
+    manager = ModelManager()
+    source = FileSystemStoragePathSource(
+        manager, loader=loader,
+        poll_wait_seconds=args.file_system_poll_wait_seconds,
+        max_num_load_retries=args.max_num_load_retries,
+        enable_warmup=args.enable_model_warmup,
+        fail_if_zero_versions_at_startup=(
+            args.fail_if_zero_versions_at_startup))
+    # TLS requires the grpcio transport (the native C++ transport is h2c)
+    transport = "grpcio" if args.ssl_config_file else "native"
     server = ModelServer(port=args.port, raw_predict=args.raw_predict,
                          device=args.device,
+                         manager=manager,
+                         transport=transport,
+                         storage_source=source,
                          servable_factory=lambda name, path:
                          default_loader(name, path, device=args.device))
     if args.grpc_socket_path:
-        server._server.add_insecure_port(f"unix://{args.grpc_socket_path}")
+        if server._native is not None:
+            server._native.add_listener(f"unix://{args.grpc_socket_path}")
+        else:
+            server._server.add_insecure_port(
+                f"unix://{args.grpc_socket_path}")
     if args.ssl_config_file:
         ssl_cfg = _parse_text_proto(args.ssl_config_file, pb.SSLConfig)
         creds = grpc.ssl_server_credentials(
@@ -105,14 +123,6 @@ def make_server(args) -> tuple:
             require_client_auth=ssl_cfg.client_verify)
         server.ssl_port = server._server.add_secure_port(
             f"127.0.0.1:{args.port + 1}", creds)
-
-    source = FileSystemStoragePathSource(
-        server.manager, loader=loader,
-        poll_wait_seconds=args.file_system_poll_wait_seconds,
-        max_num_load_retries=args.max_num_load_retries,
-        enable_warmup=args.enable_model_warmup,
-        fail_if_zero_versions_at_startup=(
-            args.fail_if_zero_versions_at_startup))
 
     configs, policies, labels = {}, {}, {}
     if args.model_config_file:

@@ -267,6 +267,10 @@ class NativeTransportServer:
         self._srv.set_echo_models(PREDICT_PATH, models)
 
     # -- lifecycle -------------------------------------------------------
+    def add_listener(self, address: str) -> None:
+        """Extra listen address (e.g. --grpc_socket_path); before start."""
+        self._srv.add_address(address)
+
     def start(self):
         self.address = self._srv.start()
         return self.address

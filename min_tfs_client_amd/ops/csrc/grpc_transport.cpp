@@ -259,11 +259,21 @@ class GrpcServer {
     }
   }
 
+  // extra listen addresses (e.g. --grpc_socket_path); call before start()
+  void add_address(const std::string& address) {
+    extra_addresses_.push_back(address);
+  }
+
   std::string start() {
-    listen_fd_ = make_listener();
+    listen_fds_.push_back(make_listener(address_, &bound_address_));
+    for (auto& extra : extra_addresses_) {
+      std::string bound;
+      listen_fds_.push_back(make_listener(extra, &bound));
+    }
     for (int i = 0; i < n_workers_; ++i)
       workers_.emplace_back([this] { worker_loop(); });
-    accept_thread_ = std::thread([this] { accept_loop(); });
+    for (int fd : listen_fds_)
+      accept_threads_.emplace_back([this, fd] { accept_loop(fd); });
     running_ = true;
     return bound_address_;
   }
@@ -327,16 +337,17 @@ class GrpcServer {
   }
   std::string address_;
   std::string bound_address_;
+  std::vector<std::string> extra_addresses_;
   int n_workers_;
-  int listen_fd_ = -1;
+  std::vector<int> listen_fds_;
   std::atomic<bool> running_{false};
   std::atomic<bool> stopping_{false};
-  std::thread accept_thread_;
+  std::vector<std::thread> accept_threads_;
   std::vector<std::thread> workers_;
   std::vector<std::thread> conn_threads_;
   std::mutex conns_mu_;
   std::vector<std::shared_ptr<Conn>> conns_;
-  std::string unix_path_;  // unlink on stop
+  std::vector<std::string> unix_paths_;  // unlink on stop
 
   std::mutex handler_mu_;
   std::unordered_map<std::string, py::object> py_handlers_;
@@ -353,9 +364,9 @@ class GrpcServer {
   std::condition_variable q_cv_;
   std::deque<Task> queue_;
 
-  int make_listener() {
-    if (address_.rfind("unix:", 0) == 0) {
-      std::string path = address_.substr(5);
+  int make_listener(const std::string& address, std::string* bound) {
+    if (address.rfind("unix:", 0) == 0) {
+      std::string path = address.substr(5);
       while (path.size() >= 2 && path[0] == '/' && path[1] == '/')
         path = path.substr(1);
       ::unlink(path.c_str());
@@ -372,16 +383,16 @@ class GrpcServer {
         ::close(fd);
         throw ConnError("bind/listen " + path + ": " + strerror(e));
       }
-      unix_path_ = path;
-      bound_address_ = address_;
+      unix_paths_.push_back(path);
+      *bound = address;
       return fd;
     }
-    auto colon = address_.rfind(':');
+    auto colon = address.rfind(':');
     std::string host =
-        colon == std::string::npos ? address_ : address_.substr(0, colon);
+        colon == std::string::npos ? address : address.substr(0, colon);
     int port = colon == std::string::npos
                    ? 0
-                   : std::atoi(address_.c_str() + colon + 1);
+                   : std::atoi(address.c_str() + colon + 1);
     if (host.empty() || host == "localhost") host = "127.0.0.1";
     int fd = ::socket(AF_INET, SOCK_STREAM, 0);
     if (fd < 0) throw ConnError("socket: " + std::string(strerror(errno)));
@@ -398,17 +409,17 @@ class GrpcServer {
         ::listen(fd, 128) < 0) {
       int e = errno;
       ::close(fd);
-      throw ConnError("bind/listen " + address_ + ": " + strerror(e));
+      throw ConnError("bind/listen " + address + ": " + strerror(e));
     }
     socklen_t alen = sizeof(addr);
     ::getsockname(fd, reinterpret_cast<sockaddr*>(&addr), &alen);
-    bound_address_ = host + ":" + std::to_string(ntohs(addr.sin_port));
+    *bound = host + ":" + std::to_string(ntohs(addr.sin_port));
     return fd;
   }
 
-  void accept_loop() {
+  void accept_loop(int listen_fd) {
     while (!stopping_) {
-      int cfd = ::accept(listen_fd_, nullptr, nullptr);
+      int cfd = ::accept(listen_fd, nullptr, nullptr);
       if (cfd < 0) {
         if (errno == EINTR) continue;
         break;  // listener closed
@@ -733,18 +744,20 @@ class GrpcServer {
   void stop_internal(bool wait) {
     bool was_running = running_.exchange(false);
     stopping_ = true;
-    if (listen_fd_ >= 0) {
-      ::shutdown(listen_fd_, SHUT_RDWR);
-      ::close(listen_fd_);
-      listen_fd_ = -1;
+    for (int fd : listen_fds_) {
+      ::shutdown(fd, SHUT_RDWR);
+      ::close(fd);
     }
+    listen_fds_.clear();
     {
       std::lock_guard<std::mutex> lk(conns_mu_);
       for (auto& c : conns_) c->mark_broken("server stopping");
     }
     q_cv_.notify_all();
     if (!was_running && !wait) return;
-    if (accept_thread_.joinable()) accept_thread_.join();
+    for (auto& th : accept_threads_)
+      if (th.joinable()) th.join();
+    accept_threads_.clear();
     {
       std::lock_guard<std::mutex> lk(conns_mu_);
       for (auto& th : conn_threads_)
@@ -755,7 +768,8 @@ class GrpcServer {
     for (auto& w : workers_)
       if (w.joinable()) w.join();
     workers_.clear();
-    if (!unix_path_.empty()) ::unlink(unix_path_.c_str());
+    for (auto& up : unix_paths_) ::unlink(up.c_str());
+    unix_paths_.clear();
     // release python handlers with the GIL held
     if (Py_IsInitialized()) {
       py::gil_scoped_acquire gil;
@@ -1152,6 +1166,8 @@ class NativeRpcError(Exception):
            py::call_guard<py::gil_scoped_release>())
       .def("register_handler", &GrpcServer::register_handler,
            py::arg("path"), py::arg("fn"))
+      .def("add_address", &GrpcServer::add_address, py::arg("address"),
+           "Add an extra listen address (call before start()).")
       .def("set_echo_models", &GrpcServer::set_echo_models, py::arg("path"),
            py::arg("models"),
            "Enable the all-C++ identity-echo fast path for `path` for the "

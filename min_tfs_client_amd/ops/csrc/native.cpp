@@ -185,6 +185,29 @@ at::Tensor tensor_from_parsed(const tfswire::ParsedTensor& t,
     case at::kChar: fill(cpu.data_ptr<int8_t>(), t.ints); break;
     case at::kByte: fill(cpu.data_ptr<uint8_t>(), t.ints); break;
     case at::kBool: fill(cpu.data_ptr<bool>(), t.ints); break;
+    case at::kComplexFloat: case at::kComplexDouble: {
+      // scomplex_val/dcomplex_val: interleaved re/im pairs
+      // (tensor.proto:59-61); repeat-last-fill repeats the last PAIR
+      const bool single = st == at::kComplexFloat;
+      int64_t pairs = single ? int64_t(t.floats.size()) / 2
+                             : int64_t(t.doubles.size()) / 2;
+      if (single) {
+        auto* dst = reinterpret_cast<float*>(cpu.data_ptr());
+        for (int64_t i = 0; i < numel; ++i) {
+          int64_t j = i < pairs ? i : (pairs > 0 ? pairs - 1 : 0);
+          dst[2 * i] = pairs > 0 ? t.floats[size_t(2 * j)] : 0.f;
+          dst[2 * i + 1] = pairs > 0 ? t.floats[size_t(2 * j + 1)] : 0.f;
+        }
+      } else {
+        auto* dst = reinterpret_cast<double*>(cpu.data_ptr());
+        for (int64_t i = 0; i < numel; ++i) {
+          int64_t j = i < pairs ? i : (pairs > 0 ? pairs - 1 : 0);
+          dst[2 * i] = pairs > 0 ? t.doubles[size_t(2 * j)] : 0.0;
+          dst[2 * i + 1] = pairs > 0 ? t.doubles[size_t(2 * j + 1)] : 0.0;
+        }
+      }
+      break;
+    }
     case at::kHalf: case at::kBFloat16: {
       // half_val holds raw uint16 bit-patterns (tensor.cc:446-464)
       auto* dst = reinterpret_cast<uint16_t*>(cpu.data_ptr());
@@ -379,16 +402,19 @@ uint64_t serialize_predict_into(py::buffer dst, bool is_request,
 uint32_t shm_wait_value(py::buffer buf, uint64_t offset, uint32_t target,
                         double timeout_s) {
   py::buffer_info info = buf.request();
-  auto* p = reinterpret_cast<volatile uint32_t*>(
+  // C++-conformant cross-process visibility: atomic acquire load (the
+  // round-1 volatile+fence worked on x86-64 but was UB-adjacent)
+  auto* p = reinterpret_cast<const std::atomic<uint32_t>*>(
       static_cast<uint8_t*>(info.ptr) + offset);
+  static_assert(sizeof(std::atomic<uint32_t>) == sizeof(uint32_t),
+                "atomic<u32> must be layout-compatible with u32");
   py::gil_scoped_release release;
   const auto deadline = std::chrono::steady_clock::now() +
       std::chrono::duration<double>(timeout_s);
   int spins = 0;
   while (true) {
-    uint32_t v = *p;
+    uint32_t v = p->load(std::memory_order_acquire);
     if (v == target) {
-      std::atomic_thread_fence(std::memory_order_acquire);
       return v;
     }
     if (std::chrono::steady_clock::now() > deadline) return 0xFFFFFFFFu;

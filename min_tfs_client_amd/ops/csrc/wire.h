@@ -332,27 +332,37 @@ inline void parse_tensor_proto(Cursor c, ParsedTensor* t) {
         t->content_bytes = uint64_t(b.end - b.p);
         break;
       }
-      case 5: {  // float_val (packed or not)
+      // float_val=5; scomplex_val=9 is interleaved re/im floats
+      // (tensor.proto:59-61) — same wire shape, dtype disambiguates
+      case 5: case 9: {
         if (wt == WT_LEN) {
           Cursor b = c.read_len_delim();
-          while (!b.done()) {
+          // bound on >=4 so a malformed 4k+r-byte payload cannot read
+          // past the submessage end; trailing bytes are an error
+          while (b.end - b.p >= 4) {
             float v; std::memcpy(&v, b.p, 4); b.p += 4;
             t->floats.push_back(v);
           }
+          if (b.p != b.end)
+            throw std::runtime_error("wire: truncated packed float field");
         } else { uint32_t raw = 0;
           if (c.end - c.p < 4) throw std::runtime_error("wire: truncated");
           std::memcpy(&raw, c.p, 4); c.p += 4;
           float v; std::memcpy(&v, &raw, 4); t->floats.push_back(v); }
         break;
       }
-      case 6: {  // double_val
+      // double_val=6; dcomplex_val=12 is interleaved re/im doubles
+      case 6: case 12: {
         if (wt == WT_LEN) {
           Cursor b = c.read_len_delim();
-          while (!b.done()) {
+          while (b.end - b.p >= 8) {
             double v; std::memcpy(&v, b.p, 8); b.p += 8;
             t->doubles.push_back(v);
           }
-        } else { if (c.end - c.p < 8) throw std::runtime_error("trunc");
+          if (b.p != b.end)
+            throw std::runtime_error("wire: truncated packed double field");
+        } else { if (c.end - c.p < 8)
+            throw std::runtime_error("wire: truncated");
           double v; std::memcpy(&v, c.p, 8); c.p += 8;
           t->doubles.push_back(v); }
         break;

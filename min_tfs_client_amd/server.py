@@ -491,11 +491,15 @@ class PredictionServiceImpl(PredictionServiceServicer):
 class ModelServiceImpl(ModelServiceServicer):
     def __init__(self, manager: ModelManager,
                  servable_factory: Optional[Callable[[str, str], Servable]]
-                 = None):
+                 = None,
+                 storage_source=None):
         self._manager = manager
         # used by HandleReloadConfigRequest to instantiate servables for
-        # config entries: (name, base_path) -> Servable
+        # config entries: (name, version_dir) -> Servable
         self._servable_factory = servable_factory
+        # preferred reload route: the FileSystemStoragePathSource, so the
+        # poller's bookkeeping and the admin API stay consistent
+        self._storage_source = storage_source
 
     def GetModelStatus(self, request, context):
         response = pb.GetModelStatusResponse()
@@ -533,7 +537,31 @@ class ModelServiceImpl(ModelServiceServicer):
             response.status.error_message = (
                 "ServerCore accepts only model_config_list")
             return response
-        wanted = {c.name: c.base_path for c in cfg.model_config_list.config}
+        entries = list(cfg.model_config_list.config)
+        if self._storage_source is not None:
+            # route through the storage source: set_models unloads removed
+            # models AND invalidates its _loaded bookkeeping, so polled
+            # models stay consistent with admin reloads
+            from .repository import VersionPolicy
+            configs = {c.name: c.base_path for c in entries}
+            policies = {
+                c.name: VersionPolicy.from_proto(
+                    c.model_version_policy
+                    if c.HasField("model_version_policy") else None)
+                for c in entries}
+            try:
+                self._storage_source.set_models(configs, policies)
+                self._storage_source.poll_once()
+                for c in entries:
+                    for label, ver in dict(c.version_labels).items():
+                        self._manager.set_version_label(c.name, label, ver)
+            except Exception as e:  # noqa: BLE001
+                response.status.error_code = pb.ErrorCode.UNKNOWN
+                response.status.error_message = str(e)
+                return response
+            response.status.error_code = pb.ErrorCode.OK
+            return response
+        wanted = {c.name: c.base_path for c in entries}
         # unload models not in the new config; load new ones via factory
         for name in self._manager.model_names():
             if name not in wanted:
@@ -542,15 +570,36 @@ class ModelServiceImpl(ModelServiceServicer):
             for name, base_path in wanted.items():
                 if name not in self._manager.model_names():
                     try:
-                        self._manager.load(name,
-                                           self._servable_factory(name,
-                                                                  base_path))
+                        self._load_via_factory(name, base_path)
                     except Exception as e:  # noqa: BLE001
                         response.status.error_code = pb.ErrorCode.UNKNOWN
                         response.status.error_message = str(e)
                         return response
         response.status.error_code = pb.ErrorCode.OK
         return response
+
+    def _load_via_factory(self, name: str, base_path: str) -> None:
+        """TF layout: base_path contains numeric version dirs; the factory
+        receives a VERSION directory. A base_path with no numeric children
+        is treated as a single version-1 directory (legacy layouts)."""
+        import os
+        versions = []
+        if os.path.isdir(base_path):
+            versions = sorted(
+                int(e) for e in os.listdir(base_path)
+                if e.isdigit() and os.path.isdir(os.path.join(base_path,
+                                                              e)))
+        if versions:
+            latest = versions[-1]
+            vdir = os.path.join(
+                base_path,
+                next(e for e in os.listdir(base_path)
+                     if e.isdigit() and int(e) == latest))
+            self._manager.load(name, self._servable_factory(name, vdir),
+                               version=latest)
+        else:
+            self._manager.load(name,
+                               self._servable_factory(name, base_path))
 
 
 # ---------------------------------------------------------------------------
@@ -643,7 +692,8 @@ class ModelServer:
                  device: str = "cpu",
                  address: Optional[str] = None,
                  shm_handshake_dir: Optional[str] = None,
-                 transport: str = "native"):
+                 transport: str = "native",
+                 storage_source=None):
         from .utils.allocator import tune_malloc
         tune_malloc()
         from .request_logging import ServerRequestLogger
@@ -662,7 +712,8 @@ class ModelServer:
                 self.manager, output_encoding, self.metrics,
                 self.request_logger)
             self.model_service = ModelServiceImpl(self.manager,
-                                                  servable_factory)
+                                                  servable_factory,
+                                                  storage_source)
             native_addr = address if address is not None \
                 else f"127.0.0.1:{port}"
             self._native = NativeTransportServer(
@@ -691,7 +742,8 @@ class ModelServer:
         self.prediction_service = PredictionServiceImpl(
             self.manager, output_encoding, self.metrics,
             self.request_logger)
-        self.model_service = ModelServiceImpl(self.manager, servable_factory)
+        self.model_service = ModelServiceImpl(self.manager, servable_factory,
+                                              storage_source)
         if raw_predict:
             from .wire import messages as _pb
             from .wire.grpc_stubs import (

@@ -252,6 +252,15 @@ class ShmListener:
 
     def _serve_conn(self, hello):
         try:
+            # only attach segments that really live in /dev/shm with our
+            # prefix: a hostile handshake file must not be able to point
+            # the server at an arbitrary server-writable file
+            for key in ("req", "resp"):
+                real = os.path.realpath(str(hello[key]))
+                if not real.startswith("/dev/shm/mi355x_"):
+                    logger.warning("shm handshake rejected: %s is not a "
+                                   "/dev/shm/mi355x_* segment", real)
+                    return
             req = _Segment(hello["req"], 0, create=False)
             resp = _Segment(hello["resp"], 0, create=False)
         except (FileNotFoundError, KeyError, ValueError):
@@ -276,8 +285,16 @@ class ShmListener:
             resp.close()
 
     def _handle_one(self, req, resp):
-        # wait until the client has consumed the previous response
-        self._native.shm_wait_value(resp.buf, _STATE_OFF, _IDLE, 10.0)
+        # wait until the client has consumed the previous response; on
+        # timeout (client stalled mid-parse) do NOT overwrite the buffer
+        # it may still be reading — drop the request instead
+        state = self._native.shm_wait_value(resp.buf, _STATE_OFF, _IDLE,
+                                            10.0)
+        if state == _TIMEOUT_SENTINEL:
+            logger.warning("shm client stalled holding the response slot; "
+                           "dropping request")
+            self._native.shm_store_value(req.buf, _STATE_OFF, _IDLE)
+            return
         rlen = _read_len(req.buf)
         try:
             # cheap span parse first: spec + filter without tensor copies
