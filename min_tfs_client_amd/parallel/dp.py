@@ -72,9 +72,22 @@ class DataParallelPredictor:
             dist.scatter(shard, None, src=0, group=self.group)
         return shard[:sizes[self.rank]]
 
+    def _gathered_row_counts(self, shard: torch.Tensor) -> List[int]:
+        """All-gathers every rank's ACTUAL output dim-0 size (a model's
+        output rows need not match its input shard rows — e.g. detection
+        models emit variable counts). Small fixed-size tensor collective,
+        so it works identically on gloo and RCCL."""
+        local = torch.tensor([shard.shape[0]], dtype=torch.int64,
+                             device=shard.device)
+        out = [torch.empty_like(local) for _ in range(self.world)]
+        dist.all_gather(out, local, group=self.group)
+        return [int(t.item()) for t in out]
+
     def _all_gather(self, shard: torch.Tensor,
                     sizes: List[int]) -> torch.Tensor:
         pad_rows = max(sizes)
+        if pad_rows == 0:
+            return shard[:0]
         padded = shard
         if shard.shape[0] < pad_rows:
             pad = torch.zeros((pad_rows - shard.shape[0],
@@ -130,8 +143,10 @@ class DataParallelPredictor:
             if not isinstance(v, torch.Tensor) or v.dim() == 0:
                 gathered[k] = v
                 continue
-            # output shard sizes can differ from inputs' only in dim 0
-            # semantics; assume row-aligned with the request shard sizes
-            in_sizes = metas[keys[0]]["sizes"]
-            gathered[k] = self._all_gather(v.to(self.device), in_sizes)
+            # gather each rank's true output row count first (round-1 bug:
+            # assuming row alignment with input shard sizes mis-gathers
+            # models whose output dim-0 differs per shard)
+            v = v.to(self.device)
+            out_sizes = self._gathered_row_counts(v)
+            gathered[k] = self._all_gather(v, out_sizes)
         return gathered

@@ -86,6 +86,79 @@ def test_dp_predict_two_ranks():
         assert status == "ok", f"rank {rank}: {status}"
 
 
+def _dp_rowchange_worker(rank, world, port, q):
+    """Servable whose output dim-0 differs from its input shard's (one
+    summary row per shard): the gather must use ACTUAL output row counts
+    (round-1 weakness #5 — row-aligned assumption mis-gathered this)."""
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        import torch.distributed as dist
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        import sys
+        root = os.path.dirname(os.path.dirname(
+            os.path.dirname(os.path.abspath(__file__))))
+        if root not in sys.path:
+            sys.path.insert(0, root)
+        from min_tfs_client_amd.parallel import DataParallelPredictor
+        from min_tfs_client_amd.server import (
+            ModelServer,
+            Servable,
+        )
+        from min_tfs_client_amd.turbo import TurboPredictClient
+
+        def reduce_fn(inputs):
+            x = inputs["x"]
+            x = torch.as_tensor(np.asarray(x))
+            # one row out per shard regardless of shard row count
+            return {"sum": x.sum(dim=0, keepdim=True)}
+
+        sock = f"unix:///tmp/dp_rc_{os.getpid()}_{rank}.sock"
+        with ModelServer(address=sock) as srv:
+            srv.manager.load("m", Servable(reduce_fn), version=1)
+            with TurboPredictClient(sock) as client:
+                dp = DataParallelPredictor(client, device="cpu")
+                full = None
+                if rank == 0:
+                    torch.manual_seed(3)
+                    full = {"x": torch.randn(7, 5)}
+                out = dp.predict("m", full)
+                # world rows: one summary row per rank
+                assert out["sum"].shape == (world, 5)
+                if rank == 0:
+                    sizes = shard_sizes(7, world)
+                    expect = torch.cat([
+                        full["x"][sum(sizes[:r]):sum(sizes[:r + 1])].sum(
+                            dim=0, keepdim=True)
+                        for r in range(world)], dim=0)
+                    assert torch.allclose(out["sum"], expect, atol=1e-5)
+        dist.destroy_process_group()
+        q.put((rank, "ok"))
+    except Exception as e:  # noqa: BLE001
+        import traceback
+        q.put((rank, f"FAIL: {e}\n{traceback.format_exc()}"))
+
+
+@pytest.mark.timeout(120)
+def test_dp_gather_with_row_count_change():
+    import socket
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    ctx = multiprocessing.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_dp_rowchange_worker, args=(r, 2, port, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=100) for _ in range(2)]
+    for p in procs:
+        p.join(timeout=30)
+    for rank, status in results:
+        assert status == "ok", f"rank {rank}: {status}"
+
+
 def _dp_shm_worker(rank, world, port, q):
     try:
         os.environ["MASTER_ADDR"] = "127.0.0.1"
