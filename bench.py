@@ -181,10 +181,19 @@ def main():
         addresses.append(address)
         readies.append(ready)
         procs.append(proc)
-    # fresh boxes: first torch import in the children can take minutes
-    for ready in readies:
-        if not ready.wait(300):
-            raise RuntimeError("bench server failed to start")
+    # fresh boxes: first torch import in the children can take minutes;
+    # poll so a crashed server child fails the bench fast instead of
+    # hanging for the full window.
+    import time as _time
+    for ready, proc in zip(readies, procs):
+        deadline = _time.monotonic() + 300
+        while not ready.wait(2):
+            if not proc.is_alive():
+                raise RuntimeError(
+                    "bench server process died during startup "
+                    f"(exitcode={proc.exitcode})")
+            if _time.monotonic() > deadline:
+                raise RuntimeError("bench server failed to start (timeout)")
     address = addresses[0] if len(addresses) == 1 else addresses
 
     # ---- client ------------------------------------------------------
