@@ -104,6 +104,12 @@ def main():
     ap.add_argument("--copy-mode", type=int, default=1,
                     help="0 = pinned-staged pipelined copies, 1 = direct "
                          "pageable hipMemcpy (A/B)")
+    ap.add_argument("--streaming", default="auto",
+                    choices=["auto", "on", "off"],
+                    help="skeleton+regions send (device DMA chunks "
+                         "overlap DATA frames; host regions zero-copy "
+                         "iovec). auto = on for the native transport "
+                         "(A/B with off)")
     ap.add_argument("--transport", default="unix",
                     choices=["unix", "tcp", "shm"],
                     help="unix/tcp = gRPC; shm = shared-memory local "
@@ -262,16 +268,21 @@ def main():
         transform = ({"images": ("nhwc", torch.float32)}
                      if args.bench_config == "fused" else None)
 
+        stream_arg = (None if args.streaming == "auto"
+                      else args.streaming == "on")
+
         def step_fn(step_inputs):
             out_dev = f"cuda:{local_rank}" if has_gpu else "cpu"
             if args.shards > 1 and transform is None:
                 return client.predict_sharded(
                     "default", step_inputs, shards=args.shards,
-                    output_device=out_dev, copy_mode=args.copy_mode)
+                    output_device=out_dev, copy_mode=args.copy_mode,
+                    streaming=stream_arg)
             return client.predict("default", step_inputs,
                                   output_device=out_dev,
                                   copy_mode=args.copy_mode,
-                                  transform=transform)
+                                  transform=transform,
+                                  streaming=stream_arg)
     else:
         from min_tfs_client_amd.client import TensorServingClient
         if args.transport == "unix":
@@ -429,6 +440,7 @@ def main():
                 "encoding": args.encoding,
                 "grpc_impl": args.grpc_impl,
                 "copy_mode": args.copy_mode,
+                "streaming": args.streaming,
                 "transport": args.transport,
                 "servers_per_rank": args.servers,
                 "gpu": has_gpu,

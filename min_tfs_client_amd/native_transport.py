@@ -100,6 +100,24 @@ def _serialize_response_py(spec, version, outputs,
     return resp.SerializeToString()
 
 
+class StreamingReply:
+    """Handler return value for the C++ server's overlapped send path:
+    the wire skeleton plus (offset, nbytes, ptr, is_device) payload
+    regions. The server sends skeleton spans from `buffer`, host regions
+    zero-copy from tensor memory, and device regions through the pooled
+    pinned-staging pipeline with the socket write as the chunk consumer —
+    hipMemcpyAsync of chunk i+1 overlaps the DATA-frame send of chunk i.
+    `keepalive` pins the region tensors until the send completes (the
+    server holds this object through the send)."""
+
+    __slots__ = ("buffer", "_wire_regions", "keepalive")
+
+    def __init__(self, buffer, regions, keepalive):
+        self.buffer = buffer
+        self._wire_regions = regions
+        self.keepalive = keepalive
+
+
 def _raw_predict_bytes_handler(manager, device: str,
                                metrics: MetricsRegistry,
                                request_logger=None,
@@ -158,6 +176,18 @@ def _raw_predict_bytes_handler(manager, device: str,
         if tensors is None:
             blob = _serialize_response_py(spec, version, outputs,
                                           use_content)
+        elif request_logger is None:
+            # streaming reply: skeleton + payload regions; the C++ server
+            # overlaps DMA (device) / sends zero-copy (host). Skipped when
+            # request logging is on — the logger needs the full bytes.
+            blob, regions, keepalive = native.serialize_predict_streaming(
+                False, spec["name"], -1 if version is None else version,
+                spec["signature_name"] or "serving_default", names,
+                tensors)
+            if regions and any(r[3] for r in regions):
+                torch.cuda.current_stream().synchronize()
+            metrics.observe_request("predict", time.perf_counter() - t0)
+            return StreamingReply(blob, regions, keepalive)
         else:
             blob = native.serialize_predict_response(
                 spec["name"], -1 if version is None else version,

@@ -35,6 +35,7 @@
 
 #include "h2core.h"
 #include "hpack.h"
+#include "staging.h"
 #include "wire.h"
 
 namespace py = pybind11;
@@ -45,6 +46,70 @@ using h2::Buf;
 using h2::Conn;
 using h2::ConnError;
 using h2::FrameHeader;
+
+// ---------------------------------------------------------------------------
+// streaming send: skeleton buffer + payload regions
+// ---------------------------------------------------------------------------
+// A streaming message is the wire skeleton (tensor_content holes unwritten)
+// plus an ascending list of payload regions. Host regions are written
+// straight from tensor memory (zero-copy send — the reference's two-slice
+// encode, grpc_tensor_coding.cc:140-248, as iovec gathers); device regions
+// are staged through the pooled pinned pipeline (staging.h) with the
+// socket write as the chunk consumer, so the hipMemcpyAsync of chunk i+1
+// overlaps the DATA-frame send of chunk i.
+struct WireRegion {
+  size_t offset;
+  size_t nbytes;
+  uintptr_t ptr;
+  bool device;
+};
+
+void write_message_with_regions(Conn& conn, uint32_t stream,
+                                const uint8_t* buf, size_t len,
+                                const std::vector<WireRegion>& regions,
+                                bool end_stream) {
+  h2::DataMessageWriter w(conn, stream, len, end_stream);
+  size_t pos = 0;
+  for (const auto& r : regions) {
+    if (r.offset < pos || r.nbytes > len || r.offset > len - r.nbytes)
+      throw ConnError("bad streaming region");
+    if (r.offset > pos) w.write(buf + pos, r.offset - pos);
+    if (!r.device) {
+      w.write(reinterpret_cast<const uint8_t*>(r.ptr), r.nbytes);
+    } else {
+      mi355x_staging::Lease lease;
+      lease.ctx->d2h(reinterpret_cast<const void*>(r.ptr), r.nbytes,
+                     [&w](const void* chunk, size_t n) {
+                       w.write(static_cast<const uint8_t*>(chunk), n);
+                     });
+    }
+    pos = r.offset + r.nbytes;
+  }
+  if (pos < len) w.write(buf + pos, len - pos);
+  w.finish();
+}
+
+// GIL must be held. Returns true when `result` is a streaming reply
+// (duck-typed on ._wire_regions), filling the skeleton buffer view and
+// region list. The buffer_info pins the skeleton; the python object itself
+// keeps the region tensors alive (.keepalive).
+bool extract_streaming_reply(py::handle result,
+                             std::unique_ptr<py::buffer_info>* info,
+                             const uint8_t** buf, size_t* len,
+                             std::vector<WireRegion>* regions) {
+  if (!py::hasattr(result, "_wire_regions")) return false;
+  py::object buffer = result.attr("buffer");
+  *info = std::make_unique<py::buffer_info>(py::buffer(buffer).request());
+  *buf = static_cast<const uint8_t*>((*info)->ptr);
+  *len = size_t((*info)->size) * size_t((*info)->itemsize);
+  for (auto item : result.attr("_wire_regions")) {
+    auto t = py::reinterpret_borrow<py::tuple>(item);
+    regions->push_back(WireRegion{
+        t[0].cast<size_t>(), t[1].cast<size_t>(),
+        uintptr_t(t[2].cast<uint64_t>()), t[3].cast<bool>()});
+  }
+  return true;
+}
 
 // ---------------------------------------------------------------------------
 // grpc status error (translated to min_tfs_client_amd NativeRpcError)
@@ -695,19 +760,28 @@ class GrpcServer {
       send_error_response(t, err_code, err_msg);
       return;
     }
-    // zero-copy send from the python result's buffer; the view pins it
+    // zero-copy send from the python result's buffer; the view pins it.
+    // Streaming replies (skeleton + payload regions) take the overlapped
+    // region path; plain buffers are sent whole.
     std::unique_ptr<py::buffer_info> info;
     const uint8_t* ptr = nullptr;
     size_t len = 0;
+    bool streaming = false;
+    std::vector<WireRegion> regions;
     {
       py::gil_scoped_acquire gil;
       try {
-        info = std::make_unique<py::buffer_info>(
-            py::buffer(result).request());
-        ptr = static_cast<const uint8_t*>(info->ptr);
-        len = size_t(info->size) * size_t(info->itemsize);
+        streaming = extract_streaming_reply(result, &info, &ptr, &len,
+                                            &regions);
+        if (!streaming) {
+          info = std::make_unique<py::buffer_info>(
+              py::buffer(result).request());
+          ptr = static_cast<const uint8_t*>(info->ptr);
+          len = size_t(info->size) * size_t(info->itemsize);
+        }
       } catch (...) {
         info.reset();
+        ptr = nullptr;
       }
     }
     if (!ptr) {
@@ -715,7 +789,17 @@ class GrpcServer {
                           "handler returned a non-buffer object");
     } else {
       try {
-        send_ok_response(t, ptr, len);
+        if (streaming) {
+          t.conn->send_headers(t.stream, make_response_headers_block(),
+                               false);
+          write_message_with_regions(*t.conn, t.stream, ptr, len, regions,
+                                     false);
+          t.conn->send_headers(t.stream, make_trailers_block(GRPC_OK, ""),
+                               true);
+          t.conn->close_send_stream(t.stream);
+        } else {
+          send_ok_response(t, ptr, len);
+        }
       } catch (...) {
         py::gil_scoped_acquire gil;
         info.reset();
@@ -808,9 +892,8 @@ class GrpcChannel {
     fail_all_pending(GRPC_UNAVAILABLE, "channel closed");
   }
 
-  // starts a unary call; returns the stream id used as a handle
-  uint32_t start_call(const std::string& path, const uint8_t* data,
-                      size_t len, double timeout_s) {
+  // opens a new stream and sends its HEADERS; shared by both call styles
+  uint32_t begin_stream(const std::string& path, double timeout_s) {
     auto p = std::make_shared<Pending>();
     uint32_t id;
     {
@@ -845,8 +928,34 @@ class GrpcChannel {
         throw RpcCallError(GRPC_UNAVAILABLE, e.what());
       }
     }
+    return id;
+  }
+
+  // starts a unary call; returns the stream id used as a handle
+  uint32_t start_call(const std::string& path, const uint8_t* data,
+                      size_t len, double timeout_s) {
+    uint32_t id = begin_stream(path, timeout_s);
     try {
       conn_->send_data_message(id, data, len, true);
+    } catch (const std::exception& e) {
+      std::lock_guard<std::mutex> lk(mu_);
+      pending_.erase(id);
+      conn_->close_send_stream(id);
+      throw RpcCallError(GRPC_UNAVAILABLE, e.what());
+    }
+    return id;
+  }
+
+  // streaming variant: the request payload is a skeleton plus regions
+  // (see write_message_with_regions) — device regions overlap DMA with
+  // the send, host regions go out zero-copy straight from tensor memory
+  uint32_t start_call_streaming(const std::string& path, const uint8_t* buf,
+                                size_t len,
+                                const std::vector<WireRegion>& regions,
+                                double timeout_s) {
+    uint32_t id = begin_stream(path, timeout_s);
+    try {
+      write_message_with_regions(*conn_, id, buf, len, regions, true);
     } catch (const std::exception& e) {
       std::lock_guard<std::mutex> lk(mu_);
       pending_.erase(id);
@@ -1195,6 +1304,58 @@ class NativeRpcError(Exception):
             return OwnedBuf(std::move(resp));
           },
           py::arg("path"), py::arg("data"), py::arg("timeout") = 0.0)
+      .def(
+          "call_streaming",
+          [](GrpcChannel& ch, const std::string& path, py::buffer data,
+             py::list region_list, double timeout) {
+            py::buffer_info info = data.request();
+            const uint8_t* ptr = static_cast<const uint8_t*>(info.ptr);
+            size_t len = size_t(info.size) * size_t(info.itemsize);
+            std::vector<WireRegion> regions;
+            regions.reserve(py::len(region_list));
+            for (auto item : region_list) {
+              auto t = py::reinterpret_borrow<py::tuple>(item);
+              regions.push_back(WireRegion{
+                  t[0].cast<size_t>(), t[1].cast<size_t>(),
+                  uintptr_t(t[2].cast<uint64_t>()), t[3].cast<bool>()});
+            }
+            py::gil_scoped_release release;
+            uint32_t id =
+                ch.start_call_streaming(path, ptr, len, regions, timeout);
+            Buf resp = ch.wait(id, timeout);
+            return OwnedBuf(std::move(resp));
+          },
+          py::arg("path"), py::arg("data"), py::arg("regions"),
+          py::arg("timeout") = 0.0,
+          "Unary call whose request payload is a skeleton buffer plus "
+          "(offset, nbytes, ptr, is_device) regions: device regions "
+          "overlap staging DMA with the send; host regions are sent "
+          "zero-copy from tensor memory. Caller must keep region memory "
+          "alive for the duration of the call.")
+      .def(
+          "start_streaming",
+          [](GrpcChannel& ch, const std::string& path, py::buffer data,
+             py::list region_list, double timeout) {
+            py::buffer_info info = data.request();
+            const uint8_t* ptr = static_cast<const uint8_t*>(info.ptr);
+            size_t len = size_t(info.size) * size_t(info.itemsize);
+            std::vector<WireRegion> regions;
+            regions.reserve(py::len(region_list));
+            for (auto item : region_list) {
+              auto t = py::reinterpret_borrow<py::tuple>(item);
+              regions.push_back(WireRegion{
+                  t[0].cast<size_t>(), t[1].cast<size_t>(),
+                  uintptr_t(t[2].cast<uint64_t>()), t[3].cast<bool>()});
+            }
+            py::gil_scoped_release release;
+            return ch.start_call_streaming(path, ptr, len, regions,
+                                           timeout);
+          },
+          py::arg("path"), py::arg("data"), py::arg("regions"),
+          py::arg("timeout") = 0.0,
+          "Streaming-region variant of start(); the send completes before "
+          "this returns (region memory may be released), the response is "
+          "collected with wait().")
       .def(
           "start",
           [](GrpcChannel& ch, const std::string& path, py::buffer data,

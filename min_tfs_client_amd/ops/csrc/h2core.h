@@ -471,6 +471,89 @@ struct Conn {
   }
 };
 
+// Streams ONE gRPC message of known total payload length as flow-controlled
+// DATA frames, fed incrementally: write() is called with consecutive byte
+// spans that must sum to exactly payload_len. This lets a producer
+// interleave generation with transmission — the transport's streaming send
+// hands each pinned-staging DMA chunk here while the NEXT chunk is still
+// copying off the device (the north star's "hipMemcpyAsync overlaps the
+// gRPC send"). END_STREAM is set on the frame that completes the message
+// iff `end_stream`.
+struct DataMessageWriter {
+  Conn& conn;
+  uint32_t stream;
+  size_t total;  // 5-byte gRPC prefix + payload
+  bool end_stream;
+  uint8_t prefix[5];
+  size_t sent = 0;  // virtual offset into (prefix || payload)
+
+  DataMessageWriter(Conn& c, uint32_t s, size_t payload_len, bool es)
+      : conn(c), stream(s), total(payload_len + 5), end_stream(es) {
+    prefix[0] = 0;  // not compressed
+    put_be32(prefix + 1, uint32_t(payload_len));
+  }
+
+  // feed the next `n` payload bytes (in order); sends them (plus the
+  // prefix, on the first call) as DATA frames respecting both windows
+  void write(const uint8_t* payload, size_t n) {
+    size_t virt_avail = (sent < 5 ? 5 - sent : 0) + n;
+    size_t consumed_payload = 0;
+    while (virt_avail > 0) {
+      size_t chunk;
+      {
+        std::unique_lock<std::mutex> lk(conn.fc_mu);
+        conn.fc_cv.wait(lk, [&] {
+          if (conn.broken) return true;
+          auto it = conn.stream_send_window.find(stream);
+          int64_t sw = it == conn.stream_send_window.end() ? 0 : it->second;
+          return conn.conn_send_window > 0 && sw > 0;
+        });
+        if (conn.broken)
+          throw ConnError("connection broken: " + conn.broken_why);
+        int64_t sw = conn.stream_send_window[stream];
+        int64_t avail =
+            conn.conn_send_window < sw ? conn.conn_send_window : sw;
+        chunk = size_t(avail);
+        if (chunk > virt_avail) chunk = virt_avail;
+        if (chunk > conn.peer_max_frame) chunk = conn.peer_max_frame;
+        conn.conn_send_window -= int64_t(chunk);
+        conn.stream_send_window[stream] -= int64_t(chunk);
+      }
+      bool last = (sent + chunk == total);
+      uint8_t fh[9];
+      put_frame_header(fh, uint32_t(chunk), F_DATA,
+                       (last && end_stream) ? FL_END_STREAM : 0, stream);
+      struct iovec iov[3];
+      int iovcnt = 0;
+      iov[iovcnt++] = {fh, 9};
+      size_t left = chunk;
+      if (sent < 5) {
+        size_t pre = 5 - sent < left ? 5 - sent : left;
+        iov[iovcnt++] = {prefix + sent, pre};
+        left -= pre;
+      }
+      if (left > 0) {
+        iov[iovcnt++] = {const_cast<uint8_t*>(payload) + consumed_payload,
+                         left};
+      }
+      {
+        std::lock_guard<std::mutex> lk(conn.write_mu);
+        writev_all(conn.fd, iov, iovcnt);
+      }
+      sent += chunk;
+      consumed_payload += left;
+      virt_avail -= chunk;
+    }
+  }
+
+  // flush a zero-payload message's prefix (write() never called)
+  void finish() {
+    if (sent < total && total == 5) write(nullptr, 0);
+  }
+
+  bool complete() const { return sent == total; }
+};
+
 // grpc-message percent coding (gRPC HTTP/2 protocol spec)
 inline std::string percent_encode(const std::string& s) {
   static const char* hex = "0123456789ABCDEF";

@@ -9,6 +9,7 @@
 
 #include <torch/extension.h>
 
+#include <algorithm>
 #include <atomic>
 #include <chrono>
 #include <cstring>
@@ -130,6 +131,80 @@ py::bytes serialize_predict(bool is_request, const std::string& model_name,
     }
   }
   return py::reinterpret_steal<py::bytes>(obj);
+}
+
+// Streaming serialize: writes only the wire SKELETON (everything except
+// large tensor_content payloads) and returns the payload locations as
+// regions, so the transport can stream them — pinned-staging DMA chunks
+// straight into DATA frames for device tensors (copy/send overlap), iovec
+// directly from tensor memory for host tensors (true zero-copy send, the
+// analogue of the reference's two-slice encode,
+// grpc_tensor_coding.cc:140-248).
+//
+// Returns (skeleton_bytes, regions, keepalive):
+//   regions  = [(offset, nbytes, ptr, is_device), ...] ascending offsets;
+//   keepalive = the contiguous tensors backing `ptr` — the caller must
+//   hold it until the send completes.
+// Host spans smaller than kStreamMinSpan are copied into the skeleton
+// (fewer iovec regions beats the copy at that size).
+constexpr size_t kStreamMinSpan = 64 * 1024;
+
+py::tuple serialize_predict_streaming(bool is_request,
+                                      const std::string& model_name,
+                                      int64_t version,
+                                      const std::string& signature,
+                                      const std::vector<std::string>& names,
+                                      const std::vector<at::Tensor>& tensors) {
+  TORCH_CHECK(names.size() == tensors.size(), "names/tensors mismatch");
+  std::vector<tfswire::TensorMeta> metas(tensors.size());
+  std::vector<at::Tensor> contig(tensors.size());
+  for (size_t i = 0; i < tensors.size(); ++i) {
+    contig[i] = tensors[i].contiguous();
+    metas[i].dtype = torch_to_tf_dtype(contig[i].scalar_type());
+    auto sizes = contig[i].sizes();
+    metas[i].shape.assign(sizes.begin(), sizes.end());
+    metas[i].content_bytes =
+        uint64_t(contig[i].numel()) * contig[i].element_size();
+    TORCH_CHECK(metas[i].content_bytes < (uint64_t(1) << 31),
+                "tensor '", names[i], "' exceeds the 2GB tensor_content "
+                "limit of the protobuf wire format");
+  }
+  auto plan = tfswire::plan_predict_message(is_request, model_name, version,
+                                            signature, names, metas);
+  PyObject* obj = PyBytes_FromStringAndSize(nullptr,
+                                            Py_ssize_t(plan.total_size));
+  if (!obj) throw std::bad_alloc();
+  auto* buf = reinterpret_cast<uint8_t*>(PyBytes_AS_STRING(obj));
+  tfswire::write_predict_message(buf, plan, is_request, model_name, version,
+                                 signature, names, metas);
+  py::list regions;
+  py::list keepalive;
+  // plan.spans arrive in tensor order; region list must ascend by offset
+  std::vector<size_t> order(contig.size());
+  for (size_t i = 0; i < order.size(); ++i) order[i] = i;
+  std::sort(order.begin(), order.end(), [&](size_t a, size_t b) {
+    return plan.spans[a].offset < plan.spans[b].offset;
+  });
+  for (size_t idx : order) {
+    const auto& span = plan.spans[idx];
+    if (span.nbytes == 0) continue;
+    const at::Tensor& t = contig[idx];
+    if (t.is_cuda()) {
+      regions.append(py::make_tuple(
+          uint64_t(span.offset), uint64_t(span.nbytes),
+          uint64_t(reinterpret_cast<uintptr_t>(t.const_data_ptr())), true));
+      keepalive.append(t);
+    } else if (span.nbytes >= kStreamMinSpan) {
+      regions.append(py::make_tuple(
+          uint64_t(span.offset), uint64_t(span.nbytes),
+          uint64_t(reinterpret_cast<uintptr_t>(t.const_data_ptr())), false));
+      keepalive.append(t);
+    } else {
+      std::memcpy(buf + span.offset, t.const_data_ptr(), span.nbytes);
+    }
+  }
+  return py::make_tuple(py::reinterpret_steal<py::bytes>(obj), regions,
+                        keepalive);
 }
 
 // ---------------------------------------------------------------------------
@@ -490,6 +565,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         },
         py::arg("model_name"), py::arg("version"), py::arg("signature"),
         py::arg("names"), py::arg("tensors"), py::arg("copy_mode") = 0);
+  m.def("serialize_predict_streaming", &serialize_predict_streaming,
+        py::arg("is_request"), py::arg("model_name"), py::arg("version"),
+        py::arg("signature"), py::arg("names"), py::arg("tensors"),
+        "Skeleton + payload-region streaming serialize: returns (bytes, "
+        "regions[(offset,nbytes,ptr,is_device)], keepalive) for the "
+        "transport's overlapped/zero-copy send path.");
   m.def("parse_predict_request",
         [](py::buffer b, const std::string& device, int copy_mode) {
           return parse_predict(b, true, device, copy_mode);

@@ -112,6 +112,19 @@ class _NativeStub:
         call_id = self._channel.start(self._path, blob, timeout or 0.0)
         return _NativeFuture(self._channel, call_id, timeout or 0.0)
 
+    @_translate_native_error
+    def call_streaming(self, blob, regions, timeout=60.0):
+        """Skeleton+regions request send (overlapped DMA for device
+        regions, zero-copy iovec for host regions)."""
+        return self._channel.call_streaming(self._path, blob, regions,
+                                            timeout or 0.0)
+
+    @_translate_native_error
+    def future_streaming(self, blob, regions, timeout=60.0):
+        call_id = self._channel.start_streaming(self._path, blob, regions,
+                                                timeout or 0.0)
+        return _NativeFuture(self._channel, call_id, timeout or 0.0)
+
 
 class TurboPredictClient:
     """Raw-bytes Predict client over the C++ codec.
@@ -207,6 +220,22 @@ class TurboPredictClient:
             model_name, -1 if model_version is None else model_version,
             signature_name, names, tensors, copy_mode)
 
+    def _serialize_streaming(self, model_name, inputs, model_version,
+                             signature_name):
+        """Skeleton + payload regions for the overlapped/zero-copy send
+        (native backend only). Returns (blob, regions, keepalive); device
+        regions are synchronized against the producing torch stream here,
+        so the transport can DMA them chunk-by-chunk into DATA frames."""
+        names = list(inputs.keys())
+        tensors = [inputs[k] for k in names]
+        blob, regions, keepalive = self._native.serialize_predict_streaming(
+            True, model_name,
+            -1 if model_version is None else model_version,
+            signature_name, names, tensors)
+        if torch is not None and any(r[3] for r in regions):
+            torch.cuda.current_stream().synchronize()
+        return blob, regions, keepalive
+
     def predict(self, model_name: str, inputs: Dict[str, "torch.Tensor"],
                 timeout: float = 60.0,
                 model_version: Optional[int] = None,
@@ -214,7 +243,8 @@ class TurboPredictClient:
                 output_device: Optional[Union[str, "torch.device"]] = None,
                 copy_mode: int = 1,
                 zero_copy: bool = False,
-                transform: Optional[Dict[str, tuple]] = None
+                transform: Optional[Dict[str, tuple]] = None,
+                streaming: Optional[bool] = None
                 ) -> Dict[str, "torch.Tensor"]:
         """One Predict round trip. ``output_device``: where response
         tensors land ("cpu" default; "cuda:N" unpacks over the staging
@@ -246,14 +276,32 @@ class TurboPredictClient:
                     elif layout == "nchw":
                         t = t.permute(0, 3, 1, 2).contiguous()
                     inputs[name] = t.to(dtype) if dtype else t
-        with trace_span("turbo.serialize", model=model_name,
-                        bytes=sum(t.numel() * t.element_size()
-                                  for t in inputs.values())):
-            blob = self.serialize_request(model_name, inputs, model_version,
-                                          signature_name, copy_mode)
-        self.metrics.observe_bytes("tx", len(blob))
-        with trace_span("turbo.rpc", bytes=len(blob)):
-            resp = self._predict(blob, timeout)
+        use_streaming = (self.backend == "native" if streaming is None
+                         else streaming)
+        if use_streaming and self.backend != "native":
+            raise ValueError("streaming=True requires backend='native'")
+        if use_streaming:
+            # skeleton + regions: DMA chunks stream into DATA frames
+            # (device), iovec from tensor memory (host) — no wire-buffer
+            # payload copy at all
+            with trace_span("turbo.serialize", model=model_name,
+                            streaming=True):
+                blob, regions, keepalive = self._serialize_streaming(
+                    model_name, inputs, model_version, signature_name)
+            self.metrics.observe_bytes("tx", len(blob))
+            with trace_span("turbo.rpc", bytes=len(blob), streaming=True):
+                resp = self._predict.call_streaming(blob, regions, timeout)
+            del keepalive
+        else:
+            with trace_span("turbo.serialize", model=model_name,
+                            bytes=sum(t.numel() * t.element_size()
+                                      for t in inputs.values())):
+                blob = self.serialize_request(model_name, inputs,
+                                              model_version,
+                                              signature_name, copy_mode)
+            self.metrics.observe_bytes("tx", len(blob))
+            with trace_span("turbo.rpc", bytes=len(blob)):
+                resp = self._predict(blob, timeout)
         self.metrics.observe_bytes("rx", len(resp))
         dev = str(output_device) if output_device is not None else "cpu"
         if zero_copy and dev == "cpu":
@@ -301,7 +349,9 @@ class TurboPredictClient:
                         model_version: Optional[int] = None,
                         signature_name: str = "",
                         output_device: Optional[str] = None,
-                        copy_mode: int = 1) -> Dict[str, "torch.Tensor"]:
+                        copy_mode: int = 1,
+                        streaming: Optional[bool] = None
+                        ) -> Dict[str, "torch.Tensor"]:
         """One logical Predict split along dim 0 into `shards` parallel
         rpcs over separate channels: transport for a single request is
         parallelized (each shard's serialize/send/recv/parse overlaps the
@@ -312,26 +362,44 @@ class TurboPredictClient:
         """
         if shards <= 1 or len(self._stubs) < 2:
             return self.predict(model_name, inputs, timeout, model_version,
-                                signature_name, output_device, copy_mode)
+                                signature_name, output_device, copy_mode,
+                                streaming=streaming)
         keys = list(inputs.keys())
         if any(inputs[k].dim() == 0 for k in keys):
             return self.predict(model_name, inputs, timeout, model_version,
-                                signature_name, output_device, copy_mode)
+                                signature_name, output_device, copy_mode,
+                                streaming=streaming)
         batch = inputs[keys[0]].shape[0]
         if any(inputs[k].shape[0] != batch for k in keys) \
                 or batch < shards:
             return self.predict(model_name, inputs, timeout, model_version,
-                                signature_name, output_device, copy_mode)
+                                signature_name, output_device, copy_mode,
+                                streaming=streaming)
         base, rem = divmod(batch, shards)
         sizes = [base + (1 if i < rem else 0) for i in range(shards)]
         futs = []
         off = 0
+        use_streaming = (self.backend == "native" if streaming is None
+                         else streaming)
+        if use_streaming and self.backend != "native":
+            raise ValueError("streaming=True requires backend='native'")
         for i, n in enumerate(sizes):
             shard = {k: inputs[k].narrow(0, off, n) for k in keys}
-            blob = self.serialize_request(model_name, shard, model_version,
-                                          signature_name, copy_mode)
-            futs.append(self._stubs[i % len(self._stubs)].future(blob,
-                                                                 timeout))
+            stub = self._stubs[i % len(self._stubs)]
+            if use_streaming:
+                # narrow() views are non-contiguous after dim 0 slicing
+                # only if stride games were played; serialize handles
+                # .contiguous() and the keepalive pins the copy through
+                # the (synchronous) send in future_streaming's start
+                blob, regions, keepalive = self._serialize_streaming(
+                    model_name, shard, model_version, signature_name)
+                futs.append(stub.future_streaming(blob, regions, timeout))
+                del keepalive
+            else:
+                blob = self.serialize_request(model_name, shard,
+                                              model_version,
+                                              signature_name, copy_mode)
+                futs.append(stub.future(blob, timeout))
             off += n
         dev = str(output_device) if output_device is not None else "cpu"
         parts = []
@@ -358,9 +426,15 @@ class TurboPredictClient:
                        copy_mode: int = 1):
         """Async variant for request pipelining: returns (grpc future,
         decode) — call decode(future.result()) to get output tensors."""
-        blob = self.serialize_request(model_name, inputs, model_version,
-                                      signature_name, copy_mode)
-        fut = self._next_stub().future(blob, timeout)
+        if self.backend == "native":
+            blob, regions, keepalive = self._serialize_streaming(
+                model_name, inputs, model_version, signature_name)
+            fut = self._next_stub().future_streaming(blob, regions, timeout)
+            del keepalive
+        else:
+            blob = self.serialize_request(model_name, inputs, model_version,
+                                          signature_name, copy_mode)
+            fut = self._next_stub().future(blob, timeout)
 
         def decode(resp_bytes, output_device="cpu"):
             _s, outputs, _ = self._native.parse_predict_response(

@@ -36,10 +36,16 @@ ext = CUDAExtension(
     },
 )
 
-# native gRPC (HTTP/2) transport: pure C++ (sockets + HPACK), no HIP
+# native gRPC (HTTP/2) transport: C++ (sockets + HPACK) + host-side HIP
+# runtime API only (staging.h pinned pipeline for the overlapped send) —
+# no kernels, so plain g++ with -lamdhip64 suffices
+_ROCM = os.environ.get("ROCM_PATH", "/opt/rocm")
 transport_ext = CppExtension(
     name="min_tfs_client_amd._transport",
     sources=[os.path.join(_CSRC, "grpc_transport.cpp")],
+    include_dirs=[os.path.join(_ROCM, "include")],
+    library_dirs=[os.path.join(_ROCM, "lib")],
+    libraries=["amdhip64"],
     extra_compile_args={"cxx": ["-O3", "-std=c++17"]},
 )
 

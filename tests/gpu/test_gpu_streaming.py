@@ -1,0 +1,150 @@
+"""Device-region streaming send on real hardware: the pinned-staging DMA
+chunks must reach the wire byte-identical to the buffered serializer, in
+both directions (client request regions, server StreamingReply regions).
+This is the DMA half of the north-star overlap ("hipMemcpyAsync overlaps
+the gRPC send"); tests/integration/test_streaming_send.py pins the
+CPU-reachable half."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from min_tfs_client_amd import _transport as T  # noqa: E402
+from min_tfs_client_amd.native_transport import StreamingReply  # noqa: E402
+from min_tfs_client_amd.ops import require_native  # noqa: E402
+from min_tfs_client_amd.server import ModelServer, Servable  # noqa: E402
+from min_tfs_client_amd.turbo import TurboPredictClient  # noqa: E402
+
+native = require_native()
+DEV = "cuda:0"
+
+
+def _inputs():
+    g = torch.Generator(device=DEV).manual_seed(11)
+    return {
+        # > 2 staging chunks (4 MiB each) to exercise the DMA pipeline
+        "images": torch.randn(64, 3, 224, 224, device=DEV, generator=g),
+        "small": torch.randn(4, 4, device=DEV, generator=g),
+        "ids": torch.randint(0, 30000, (128, 512), dtype=torch.int32,
+                             device=DEV, generator=g),
+    }
+
+
+def _streaming(inputs, is_request=True):
+    names = list(inputs.keys())
+    tensors = [inputs[k] for k in names]
+    parts = native.serialize_predict_streaming(
+        is_request, "m", 1, "serving_default", names, tensors)
+    torch.cuda.current_stream().synchronize()
+    return parts
+
+
+def _buffered(inputs, is_request=True):
+    names = list(inputs.keys())
+    tensors = [inputs[k] for k in names]
+    fn = (native.serialize_predict_request if is_request
+          else native.serialize_predict_response)
+    return fn("m", 1, "serving_default", names, tensors, 1)
+
+
+def test_device_regions_detected():
+    blob, regions, keepalive = _streaming(_inputs())
+    assert len(regions) == 3  # every cuda tensor becomes a device region
+    assert all(r[3] for r in regions)
+
+
+def test_client_device_streaming_bytes_identical():
+    srv = T.GrpcServer("127.0.0.1:0", 2)
+    received = []
+    srv.register_handler("/t.S/Echo",
+                         lambda v: received.append(bytes(v)) or b"ok")
+    addr = srv.start()
+    try:
+        inputs = _inputs()
+        blob, regions, keepalive = _streaming(inputs)
+        ch = T.GrpcChannel(addr)
+        try:
+            assert bytes(ch.call_streaming("/t.S/Echo", blob,
+                                           list(regions), 60.0)) == b"ok"
+        finally:
+            ch.close()
+        assert received[0] == _buffered(inputs)
+    finally:
+        srv.stop()
+
+
+def test_server_device_streaming_reply_bytes_identical():
+    outputs = _inputs()
+    expect = _buffered(outputs, is_request=False)
+
+    def handler(view):
+        blob, regions, keepalive = _streaming(outputs, is_request=False)
+        return StreamingReply(blob, list(regions), keepalive)
+
+    srv = T.GrpcServer("127.0.0.1:0", 2)
+    srv.register_handler("/t.S/Pred", handler)
+    addr = srv.start()
+    try:
+        ch = T.GrpcChannel(addr)
+        try:
+            got = bytes(ch.call("/t.S/Pred", b"x", 60.0))
+        finally:
+            ch.close()
+        assert got == expect
+    finally:
+        srv.stop()
+
+
+def test_end_to_end_gpu_streaming_predict(tmp_path):
+    """Full turbo round trip with device tensors through both streaming
+    paths (client request + server non-identity StreamingReply)."""
+    sock = f"unix://{tmp_path}/stream.sock"
+    with ModelServer(address=sock, raw_predict=True, device=DEV) as srv:
+        srv.manager.load(
+            "double", Servable(lambda d: {k: v * 2 for k, v in d.items()}),
+            version=1)
+        with TurboPredictClient(srv.address, backend="native") as c:
+            x = torch.randn(64, 3, 224, 224, device=DEV)
+            out = c.predict("double", {"images": x}, output_device=DEV,
+                            timeout=60)
+            assert out["images"].is_cuda
+            assert torch.equal(out["images"], x * 2)
+
+
+def test_concurrent_device_streaming():
+    """8 concurrent streaming sends: each leases its own staging context
+    (pooled pipeline) — results must not cross-corrupt."""
+    import threading
+
+    srv = T.GrpcServer("127.0.0.1:0", 8)
+    srv.register_handler("/t.S/Echo", lambda v: bytes(v))
+    addr = srv.start()
+    try:
+        payloads = []
+        for i in range(4):
+            t = torch.full((24, 3, 224, 224), float(i + 1), device=DEV)
+            blob, regions, keep = _streaming({"x": t})
+            expect = _buffered({"x": t})
+            payloads.append((blob, list(regions), keep, expect))
+        ch = T.GrpcChannel(addr)
+        errs = []
+
+        def worker(idx):
+            blob, regions, keep, expect = payloads[idx % len(payloads)]
+            try:
+                for _ in range(3):
+                    r = ch.call_streaming("/t.S/Echo", blob, regions, 60.0)
+                    if bytes(r) != expect:
+                        errs.append(f"mismatch idx={idx}")
+            except Exception as e:  # noqa: BLE001
+                errs.append(repr(e))
+
+        ts = [threading.Thread(target=worker, args=(i,)) for i in range(8)]
+        for t in ts:
+            t.start()
+        for t in ts:
+            t.join()
+        ch.close()
+        assert errs == []
+    finally:
+        srv.stop()
