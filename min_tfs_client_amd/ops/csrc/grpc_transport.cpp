@@ -19,6 +19,7 @@
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
+#include <algorithm>
 #include <atomic>
 #include <chrono>
 #include <condition_variable>
@@ -270,7 +271,18 @@ inline bool echo_eligible(const tfswire::ParsedPredict& req,
   return true;
 }
 
-inline Buf build_echo_response(const tfswire::ParsedPredict& req) {
+// Zero-copy echo: only the response SKELETON is materialized; payload
+// spans >= 64KB stay in the request buffer and are sent as host regions
+// (iovec straight from the request bytes). Small spans are copied into
+// the skeleton — fewer frames beats the copy at that size. The caller
+// must keep the request buffer alive through the send (Task::msg does).
+struct EchoResponse {
+  Buf skeleton;
+  std::vector<WireRegion> regions;
+};
+
+inline EchoResponse build_echo_response(const tfswire::ParsedPredict& req) {
+  constexpr size_t kEchoMinRegion = 64 * 1024;
   std::vector<std::string> names;
   std::vector<tfswire::TensorMeta> metas;
   std::vector<const uint8_t*> payloads;
@@ -293,14 +305,28 @@ inline Buf build_echo_response(const tfswire::ParsedPredict& req) {
   auto plan = tfswire::plan_predict_message(false, req.model_spec.name,
                                             req.model_spec.version, sig,
                                             names, metas);
-  Buf out(plan.total_size);
-  out.len = plan.total_size;
-  tfswire::write_predict_message(out.p, plan, false, req.model_spec.name,
+  EchoResponse r;
+  r.skeleton = Buf(plan.total_size);
+  r.skeleton.len = plan.total_size;
+  tfswire::write_predict_message(r.skeleton.p, plan, false,
+                                 req.model_spec.name,
                                  req.model_spec.version, sig, names, metas);
-  for (size_t i = 0; i < payloads.size(); ++i)
-    std::memcpy(out.p + plan.spans[i].offset, payloads[i],
-                plan.spans[i].nbytes);
-  return out;
+  // spans are emitted in tensor order; offsets ascend within the message
+  for (size_t i = 0; i < payloads.size(); ++i) {
+    const auto& span = plan.spans[i];
+    if (span.nbytes >= kEchoMinRegion) {
+      r.regions.push_back(WireRegion{
+          span.offset, span.nbytes,
+          reinterpret_cast<uintptr_t>(payloads[i]), false});
+    } else if (span.nbytes > 0) {
+      std::memcpy(r.skeleton.p + span.offset, payloads[i], span.nbytes);
+    }
+  }
+  std::sort(r.regions.begin(), r.regions.end(),
+            [](const WireRegion& a, const WireRegion& b) {
+              return a.offset < b.offset;
+            });
+  return r;
 }
 
 // ===========================================================================
@@ -701,12 +727,19 @@ class GrpcServer {
               tfswire::parse_predict_message(t.msg.p, t.msg.len, true);
           if (echo_eligible(parsed, *models)) {
             auto t0 = std::chrono::steady_clock::now();
-            Buf resp = build_echo_response(parsed);
-            send_ok_response(t, resp.p, resp.len);
+            EchoResponse resp = build_echo_response(parsed);
+            t.conn->send_headers(t.stream, make_response_headers_block(),
+                                 false);
+            write_message_with_regions(*t.conn, t.stream, resp.skeleton.p,
+                                       resp.skeleton.len, resp.regions,
+                                       false);
+            t.conn->send_headers(t.stream,
+                                 make_trailers_block(GRPC_OK, ""), true);
+            t.conn->close_send_stream(t.stream);
             double secs = std::chrono::duration<double>(
                               std::chrono::steady_clock::now() - t0)
                               .count();
-            record_stats(t.path, secs, t.msg.len, resp.len);
+            record_stats(t.path, secs, t.msg.len, resp.skeleton.len);
             return;
           }
         } catch (const RpcCallError& e) {
