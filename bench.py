@@ -261,8 +261,10 @@ def main():
         from min_tfs_client_amd.turbo import TurboPredictClient
         client = TurboPredictClient(
             address,
-            num_channels=max(min(args.pipeline, 8), args.shards,
-                             args.servers),
+            num_channels=min(8, max(min(args.pipeline, 8), args.shards,
+                                    args.servers,
+                                    args.shards * (2 if args.pipeline > 1
+                                                   else 1))),
             backend=args.grpc_impl)
 
         transform = ({"images": ("nhwc", torch.float32)}
@@ -344,19 +346,39 @@ def main():
 
     if args.pipeline > 1 and world_size > 1:
         raise SystemExit("--pipeline > 1 is single-rank only")
-    if args.pipeline > 1 and args.shards > 1:
-        raise SystemExit("--pipeline and --shards are mutually exclusive; "
-                         "pass --shards 1 with --pipeline")
     if args.pipeline > 1 and args.encoding != "turbo":
         raise SystemExit("--pipeline > 1 requires --encoding turbo")
 
     def run_pipelined(nsteps):
         """Sliding window of `pipeline` in-flight requests; returns
-        per-request submit->complete latencies."""
+        per-request submit->complete latencies. With --shards > 1 each
+        in-flight request is itself shard-parallel (thread pool of
+        `pipeline` sharded calls)."""
         out_dev = f"cuda:{local_rank}" if has_gpu else "cpu"
         lat_local = []
         inflight = []
         submitted = 0
+        if args.shards > 1:
+            from concurrent.futures import ThreadPoolExecutor
+            pool = ThreadPoolExecutor(max_workers=args.pipeline)
+            try:
+                while submitted < min(args.pipeline, nsteps):
+                    inflight.append((pool.submit(step_fn, inputs),
+                                     time.perf_counter()))
+                    submitted += 1
+                done = 0
+                while done < nsteps:
+                    fut, ts = inflight.pop(0)
+                    fut.result()
+                    lat_local.append(time.perf_counter() - ts)
+                    done += 1
+                    if submitted < nsteps:
+                        inflight.append((pool.submit(step_fn, inputs),
+                                         time.perf_counter()))
+                        submitted += 1
+            finally:
+                pool.shutdown(wait=True)
+            return lat_local
         while submitted < min(args.pipeline, nsteps):
             fut, dec = client.predict_future("default", inputs)
             inflight.append((fut, dec, time.perf_counter()))

@@ -139,6 +139,86 @@ struct OwnedBuf {
 };
 
 // ---------------------------------------------------------------------------
+// pinned receive pool
+// ---------------------------------------------------------------------------
+// Large gRPC messages (tensor payloads) are read from the socket into
+// POOLED PINNED buffers, so the subsequent H2D unpack to the GPU is a
+// true DMA at link speed instead of a pageable staged copy. Buffers
+// recycle through a size-classed freelist (hipHostMalloc is ~ms-scale, so
+// per-message allocation would erase the win); cached bytes are capped.
+// On hosts without a GPU (or if pinning fails) callers fall back to
+// malloc transparently.
+class PinnedPool {
+ public:
+  static PinnedPool& instance() {
+    static PinnedPool* pool = new PinnedPool();  // leaked: outlives HIP
+    return *pool;
+  }
+
+  static constexpr size_t kMinPinned = 1u << 20;    // pool only >=1MB
+  static constexpr size_t kMaxCached = 512u << 20;  // freelist cap
+
+  bool available() {
+    int state = state_.load(std::memory_order_acquire);
+    if (state == 0) {
+      int n = 0;
+      bool ok = hipGetDeviceCount(&n) == hipSuccess && n > 0;
+      state = ok ? 1 : -1;
+      state_.store(state, std::memory_order_release);
+    }
+    return state > 0;
+  }
+
+  // rounded-up pinned buffer or nullptr (caller falls back to malloc)
+  uint8_t* get(size_t n, size_t* cap_out) {
+    if (n < kMinPinned || !available()) return nullptr;
+    size_t cap = size_class(n);
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      auto& fl = free_[cap];
+      if (!fl.empty()) {
+        uint8_t* p = fl.back();
+        fl.pop_back();
+        cached_ -= cap;
+        *cap_out = cap;
+        return p;
+      }
+    }
+    void* p = nullptr;
+    if (hipHostMalloc(&p, cap) != hipSuccess) return nullptr;
+    *cap_out = cap;
+    return static_cast<uint8_t*>(p);
+  }
+
+  void put(uint8_t* p, size_t cap) {
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      if (cached_ + cap <= kMaxCached) {
+        free_[cap].push_back(p);
+        cached_ += cap;
+        return;
+      }
+    }
+    (void)hipHostFree(p);
+  }
+
+  static void deleter(void* ctx, uint8_t* p, size_t cap) {
+    static_cast<PinnedPool*>(ctx)->put(p, cap);
+  }
+
+ private:
+  static size_t size_class(size_t n) {
+    size_t cap = kMinPinned;
+    while (cap < n) cap <<= 1;
+    return cap;
+  }
+  std::atomic<int> state_{0};  // 0 unknown, 1 available, -1 unavailable
+  std::mutex mu_;
+  std::map<size_t, std::vector<uint8_t*>> free_;
+  size_t cached_ = 0;
+};
+
+// ---------------------------------------------------------------------------
 // in-flight message assembly (shared shape between server request streams
 // and client response streams)
 // ---------------------------------------------------------------------------
@@ -163,7 +243,13 @@ struct MsgAssembly {
             throw RpcCallError(GRPC_UNIMPLEMENTED,
                                "compressed gRPC messages not supported");
           uint32_t len = h2::be32(prefix + 1);
-          msg.alloc(len);
+          size_t cap = 0;
+          auto& pool = PinnedPool::instance();
+          uint8_t* pinned = pool.get(len, &cap);
+          if (pinned != nullptr)
+            msg.adopt(pinned, cap, &PinnedPool::deleter, &pool);
+          else
+            msg.alloc(len);
           msg.len = len;
           have_len = true;
         }

@@ -74,28 +74,52 @@ struct Buf {
   uint8_t* p = nullptr;
   size_t len = 0;
   size_t cap = 0;
+  // custom deallocation (e.g. return pinned memory to a pool); null =>
+  // plain free()
+  void (*deleter)(void* ctx, uint8_t* p, size_t cap) = nullptr;
+  void* deleter_ctx = nullptr;
   Buf() = default;
   explicit Buf(size_t n) { alloc(n); }
   Buf(const Buf&) = delete;
   Buf& operator=(const Buf&) = delete;
-  Buf(Buf&& o) noexcept : p(o.p), len(o.len), cap(o.cap) {
+  Buf(Buf&& o) noexcept
+      : p(o.p), len(o.len), cap(o.cap), deleter(o.deleter),
+        deleter_ctx(o.deleter_ctx) {
     o.p = nullptr; o.len = 0; o.cap = 0;
+    o.deleter = nullptr; o.deleter_ctx = nullptr;
   }
   Buf& operator=(Buf&& o) noexcept {
     if (this != &o) {
-      free(p);
+      release();
       p = o.p; len = o.len; cap = o.cap;
+      deleter = o.deleter; deleter_ctx = o.deleter_ctx;
       o.p = nullptr; o.len = 0; o.cap = 0;
+      o.deleter = nullptr; o.deleter_ctx = nullptr;
     }
     return *this;
   }
-  ~Buf() { free(p); }
+  ~Buf() { release(); }
+  void release() {
+    if (p != nullptr) {
+      if (deleter) deleter(deleter_ctx, p, cap);
+      else free(p);
+    }
+    p = nullptr; len = 0; cap = 0;
+    deleter = nullptr; deleter_ctx = nullptr;
+  }
   void alloc(size_t n) {
-    free(p);
+    release();
     p = static_cast<uint8_t*>(malloc(n ? n : 1));
     if (!p) throw std::bad_alloc();
     cap = n;
     len = 0;
+  }
+  // take ownership of externally allocated memory with its deleter
+  void adopt(uint8_t* ptr, size_t capacity,
+             void (*del)(void*, uint8_t*, size_t), void* ctx) {
+    release();
+    p = ptr; cap = capacity; len = 0;
+    deleter = del; deleter_ctx = ctx;
   }
 };
 
