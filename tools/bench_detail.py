@@ -67,6 +67,15 @@ def main():
         a[:] = b
     results["host_memcpy_ms"] = timeit(host_copy)
 
+    # 4b. streaming serialize: skeleton only, payloads stay in place
+    def ser_stream():
+        parts = native.serialize_predict_streaming(True, "m", -1, "",
+                                                   ["x"], [x])
+        if has_gpu:
+            torch.cuda.current_stream().synchronize()
+        return parts
+    results["serialize_streaming_skeleton_ms"] = timeit(ser_stream)
+
     # 5. raw gRPC round trip (pre-built blob, decode skipped)
     sock = f"unix:///tmp/mi355x_detail_{os.getpid()}.sock"
     with ModelServer(address=sock, raw_predict=True) as srv:
@@ -74,13 +83,22 @@ def main():
         client = TurboPredictClient(sock)
         rpc = client._predict
         results["grpc_rtt_prebuilt_ms"] = timeit(lambda: rpc(blob, 30))
+        if client.backend == "native":
+            sb, sregions, skeep = client._serialize_streaming(
+                "m", {"x": x}, None, "")
+            results["grpc_rtt_streaming_prebuilt_ms"] = timeit(
+                lambda: rpc.call_streaming(sb, list(sregions), 30))
         # tiny request for the protocol floor
         tiny = native.serialize_predict_request(
             "m", -1, "", ["x"], [torch.zeros(1)], 0)
         results["grpc_rtt_tiny_ms"] = timeit(lambda: rpc(tiny, 30))
-        # full client predict
+        # full client predict, streaming vs buffered send
         results["predict_full_ms"] = timeit(
             lambda: client.predict("m", {"x": x}, output_device=dev), reps=20)
+        if client.backend == "native":
+            results["predict_full_buffered_ms"] = timeit(
+                lambda: client.predict("m", {"x": x}, output_device=dev,
+                                       streaming=False), reps=20)
         client.close()
 
     print(json.dumps(results, indent=1))
