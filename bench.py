@@ -130,15 +130,20 @@ def main():
                          "gRPC server caps ~9 GB/s per process, so >1 "
                          "raises the per-GPU ceiling (requests spread "
                          "across instances)")
-    ap.add_argument("--pipeline", type=int, default=1,
+    ap.add_argument("--pipeline", type=int, default=-1,
                     help="in-flight requests per rank (1 = sequential; "
                          ">1 overlaps serialize/transport/parse of "
-                         "consecutive requests)")
+                         "consecutive requests). Default: 4 single-rank "
+                         "over gRPC (measured peak with shards 4), 1 "
+                         "multi-rank/shm")
     args = ap.parse_args()
     if args.shards < 0:
         args.shards = 1 if args.transport == "shm" else 4
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    if args.pipeline < 0:
+        args.pipeline = (4 if world_size == 1 and args.transport != "shm"
+                         and args.encoding == "turbo" else 1)
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     n_gpus = max(args.gpus, world_size)
