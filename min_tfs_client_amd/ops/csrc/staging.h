@@ -25,8 +25,11 @@
 
 #include <hip/hip_runtime_api.h>
 
+#include <chrono>
 #include <condition_variable>
 #include <cstdint>
+#include <cstdio>
+#include <cstdlib>
 #include <cstring>
 #include <mutex>
 #include <stdexcept>
@@ -78,9 +81,19 @@ struct Ctx {
   // device -> consumer(pinned_chunk, len), pipelined 2-deep: while the
   // consumer drains chunk i, chunk i+1 is already DMAing into the other
   // pinned buffer.
+  //
+  // Overlap proof: with MI355X_STAGING_TRACE=<path> set, each d2h call
+  // appends one JSON line with per-chunk {wait_ms, consume_ms} — wait_ms
+  // is the host-visible time hipEventSynchronize blocked before the
+  // chunk's DMA was done. After chunk 0, wait_ms ~ 0 means the DMA of
+  // chunk i+1 fully overlapped the consumer (socket write / memcpy) of
+  // chunk i; a non-overlapped pipeline would wait the full chunk DMA
+  // time (~70 us at 4 MiB) on every chunk.
   template <typename Consume>
   void d2h(const void* src_dev, size_t nbytes, Consume&& consume) {
     size_t nchunks = (nbytes + kChunk - 1) / kChunk;
+    const char* trace_path = std::getenv("MI355X_STAGING_TRACE");
+    std::vector<double> wait_ms, consume_ms;
     size_t issued = 0;
     for (size_t c = 0; c < (nchunks < 2 ? nchunks : 2); ++c) {
       size_t off = c * kChunk;
@@ -94,8 +107,18 @@ struct Ctx {
     for (size_t c = 0; c < nchunks; ++c) {
       size_t off = c * kChunk;
       size_t len = nbytes - off < kChunk ? nbytes - off : kChunk;
+      std::chrono::steady_clock::time_point t0, t1, t2;
+      if (trace_path) t0 = std::chrono::steady_clock::now();
       MI355X_STAGING_CHECK(hipEventSynchronize(evt[c & 1]));
+      if (trace_path) t1 = std::chrono::steady_clock::now();
       consume(static_cast<const void*>(buf[c & 1]), len);
+      if (trace_path) {
+        t2 = std::chrono::steady_clock::now();
+        wait_ms.push_back(
+            std::chrono::duration<double, std::milli>(t1 - t0).count());
+        consume_ms.push_back(
+            std::chrono::duration<double, std::milli>(t2 - t1).count());
+      }
       if (issued < nchunks) {
         size_t noff = issued * kChunk;
         size_t nlen = nbytes - noff < kChunk ? nbytes - noff : kChunk;
@@ -104,6 +127,22 @@ struct Ctx {
             hipMemcpyDeviceToHost, stream));
         MI355X_STAGING_CHECK(hipEventRecord(evt[issued & 1], stream));
         ++issued;
+      }
+    }
+    if (trace_path && !wait_ms.empty()) {
+      // one JSON line per d2h call; appends race-free enough via O_APPEND
+      // line writes (trace is a diagnostic, not a hot-path feature)
+      FILE* f = std::fopen(trace_path, "a");
+      if (f) {
+        std::fprintf(f, "{\"nbytes\": %zu, \"chunks\": %zu, \"wait_ms\": [",
+                     nbytes, nchunks);
+        for (size_t i = 0; i < wait_ms.size(); ++i)
+          std::fprintf(f, "%s%.4f", i ? "," : "", wait_ms[i]);
+        std::fprintf(f, "], \"consume_ms\": [");
+        for (size_t i = 0; i < consume_ms.size(); ++i)
+          std::fprintf(f, "%s%.4f", i ? "," : "", consume_ms[i]);
+        std::fprintf(f, "]}\n");
+        std::fclose(f);
       }
     }
   }

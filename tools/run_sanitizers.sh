@@ -9,3 +9,7 @@ g++ -std=c++17 -O1 -g -fsanitize=address,undefined -fno-omit-frame-pointer \
     tests/cpp/wire_test.cpp -o build/sanitize/wire_test_asan
 ./build/sanitize/wire_test_asan
 echo "ASAN+UBSAN wire test passed"
+g++ -std=c++17 -O1 -g -fsanitize=address,undefined -fno-omit-frame-pointer \
+    tests/cpp/h2_test.cpp -o build/sanitize/h2_test_asan -lpthread
+./build/sanitize/h2_test_asan
+echo "ASAN+UBSAN h2/hpack test passed"
