@@ -214,7 +214,8 @@ class NativeTransportServer:
                  address: str, device: str = "cpu",
                  metrics: Optional[MetricsRegistry] = None,
                  request_logger=None, max_workers: int = 16,
-                 output_encoding: str = "tensor_content"):
+                 output_encoding: str = "tensor_content",
+                 profiler_service=None):
         from . import _transport as T
         self._T = T
         self.manager = manager
@@ -249,6 +250,14 @@ class NativeTransportServer:
             _MS + "HandleReloadConfigRequest",
             _proto_handler(ms.HandleReloadConfigRequest,
                            pb.ReloadConfigRequest))
+        if profiler_service is not None:
+            _PROF = "/tensorflow.ProfilerService/"
+            self._srv.register_handler(
+                _PROF + "Profile",
+                _proto_handler(profiler_service.Profile, pb.ProfileRequest))
+            self._srv.register_handler(
+                _PROF + "Monitor",
+                _proto_handler(profiler_service.Monitor, pb.MonitorRequest))
         self._request_logger = request_logger
         if request_logger is not None and hasattr(request_logger,
                                                   "subscribe"):

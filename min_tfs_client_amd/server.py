@@ -22,6 +22,7 @@ Error messages follow predict_util.cc:65-85 shapes for drop-in UX.
 """
 from __future__ import annotations
 
+import os
 import threading
 import time
 from concurrent import futures
@@ -46,7 +47,9 @@ from .wire import messages as pb
 from .wire.grpc_stubs import (
     ModelServiceServicer,
     PredictionServiceServicer,
+    ProfilerServiceServicer,
     add_ModelServiceServicer_to_server,
+    add_ProfilerServiceServicer_to_server,
     add_PredictionServiceServicer_to_server,
 )
 
@@ -674,6 +677,71 @@ def _raw_predict_handler(manager: ModelManager, device: str,
         response_serializer=identity)
 
 
+class ProfilerServiceImpl(ProfilerServiceServicer):
+    """tensorflow.ProfilerService analogue — the reference model server
+    registers this service alongside Model/Prediction
+    (reference server.cc:324,339; profiler_service.proto:12-17).
+
+    ``Profile`` captures ``duration_ms`` of in-process spans from the
+    chrome-trace Tracer (utils/tracing.py) and returns them as a
+    ``trace_viewer.json`` tool_data payload; ``Monitor`` returns a
+    human-readable metrics snapshot (level 2 adds per-method latency
+    quantiles), matching the proto's "properly formatted string data"
+    contract."""
+
+    def __init__(self, metrics: MetricsRegistry):
+        self.metrics = metrics
+
+    def Profile(self, request, context):
+        import tempfile
+
+        from .utils.tracing import Tracer
+        from .wire import messages as pb
+        tracer = Tracer.get()
+        tracer.clear()
+        tracer.start()
+        dur_ms = request.duration_ms or 1000
+        time.sleep(min(dur_ms, 60_000) / 1000.0)
+        tracer.stop()
+        fd, path = tempfile.mkstemp(suffix=".json")
+        os.close(fd)
+        try:
+            n = tracer.export(path)
+            with open(path, "rb") as f:
+                data = f.read()
+        finally:
+            os.unlink(path)
+        resp = pb.ProfileResponse()
+        resp.empty_trace = n == 0
+        td = resp.tool_data.add()
+        td.name = "trace_viewer.json"
+        td.data = data
+        return resp
+
+    def Monitor(self, request, context):
+        from .wire import messages as pb
+        if request.duration_ms:
+            time.sleep(min(request.duration_ms, 10_000) / 1000.0)
+        lines = []
+        if request.timestamp:
+            lines.append(f"timestamp: {time.time():.3f}")
+        for name, value in sorted(self.metrics.counters().items()):
+            lines.append(f"{name}: {value}")
+        if request.monitoring_level >= 2:
+            with self.metrics._lock:
+                methods = sorted(self.metrics._latency)
+            for method in methods:
+                q = self.metrics.latency_quantiles(method)
+                if not q:
+                    continue
+                lines.append(f"latency[{method}]: " + ", ".join(
+                    f"{k}={v * 1e3:.3f}ms" if k != "count" else f"count={v}"
+                    for k, v in q.items()))
+        resp = pb.MonitorResponse()
+        resp.data = "\n".join(lines) + "\n"
+        return resp
+
+
 class ModelServer:
     """Build-and-start wrapper (Server::BuildAndStart analogue,
     reference server.cc:291-339).
@@ -714,6 +782,7 @@ class ModelServer:
             self.model_service = ModelServiceImpl(self.manager,
                                                   servable_factory,
                                                   storage_source)
+            self.profiler_service = ProfilerServiceImpl(self.metrics)
             native_addr = address if address is not None \
                 else f"127.0.0.1:{port}"
             self._native = NativeTransportServer(
@@ -721,7 +790,8 @@ class ModelServer:
                 native_addr, device=device, metrics=self.metrics,
                 request_logger=self.request_logger,
                 max_workers=max_workers,
-                output_encoding=output_encoding)
+                output_encoding=output_encoding,
+                profiler_service=self.profiler_service)
             self.shm_listener = None
             if shm_handshake_dir:
                 from .shm import ShmListener
@@ -744,6 +814,9 @@ class ModelServer:
             self.request_logger)
         self.model_service = ModelServiceImpl(self.manager, servable_factory,
                                               storage_source)
+        self.profiler_service = ProfilerServiceImpl(self.metrics)
+        add_ProfilerServiceServicer_to_server(self.profiler_service,
+                                              self._server)
         if raw_predict:
             from .wire import messages as _pb
             from .wire.grpc_stubs import (

@@ -141,6 +141,52 @@ def add_ModelServiceServicer_to_server(servicer, server):
     server.add_generic_rpc_handlers((generic_handler,))
 
 
+_PROF = "/tensorflow.ProfilerService/"
+
+
+class ProfilerServiceStub:
+    """Client stub; 2 rpcs (reference registers ProfilerService alongside
+    Model/Prediction — server.cc:324,339; profiler_service.proto:12-17)."""
+
+    def __init__(self, channel: grpc.Channel):
+        self.Profile = channel.unary_unary(
+            _PROF + "Profile",
+            request_serializer=pb.ProfileRequest.SerializeToString,
+            response_deserializer=pb.ProfileResponse.FromString)
+        self.Monitor = channel.unary_unary(
+            _PROF + "Monitor",
+            request_serializer=pb.MonitorRequest.SerializeToString,
+            response_deserializer=pb.MonitorResponse.FromString)
+
+
+class ProfilerServiceServicer:
+    def Profile(self, request, context):
+        context.set_code(grpc.StatusCode.UNIMPLEMENTED)
+        context.set_details("Method not implemented!")
+        raise NotImplementedError("Method not implemented!")
+
+    def Monitor(self, request, context):
+        context.set_code(grpc.StatusCode.UNIMPLEMENTED)
+        context.set_details("Method not implemented!")
+        raise NotImplementedError("Method not implemented!")
+
+
+def add_ProfilerServiceServicer_to_server(servicer, server):
+    rpc_method_handlers = {
+        "Profile": grpc.unary_unary_rpc_method_handler(
+            servicer.Profile,
+            request_deserializer=pb.ProfileRequest.FromString,
+            response_serializer=pb.ProfileResponse.SerializeToString),
+        "Monitor": grpc.unary_unary_rpc_method_handler(
+            servicer.Monitor,
+            request_deserializer=pb.MonitorRequest.FromString,
+            response_serializer=pb.MonitorResponse.SerializeToString),
+    }
+    generic_handler = grpc.method_handlers_generic_handler(
+        "tensorflow.ProfilerService", rpc_method_handlers)
+    server.add_generic_rpc_handlers((generic_handler,))
+
+
 # Raw-bytes stubs for the zero-(re)serialize hot path: the C++ codec emits
 # finished request bytes; identity (de)serializers hand them to grpc's C core
 # untouched, skipping python-protobuf entirely (design per the reference's

@@ -33,6 +33,15 @@ TensorInfo = _s.get_message_class("tensorflow.TensorInfo")
 SignatureDef = _s.get_message_class("tensorflow.SignatureDef")
 AssetFileDef = _s.get_message_class("tensorflow.AssetFileDef")
 
+# --- tensorflow profiler service --------------------------------------------
+ProfileOptions = _s.get_message_class("tensorflow.ProfileOptions")
+ToolRequestOptions = _s.get_message_class("tensorflow.ToolRequestOptions")
+ProfileRequest = _s.get_message_class("tensorflow.ProfileRequest")
+ProfileToolData = _s.get_message_class("tensorflow.ProfileToolData")
+ProfileResponse = _s.get_message_class("tensorflow.ProfileResponse")
+MonitorRequest = _s.get_message_class("tensorflow.MonitorRequest")
+MonitorResponse = _s.get_message_class("tensorflow.MonitorResponse")
+
 # --- tensorflow_serving.apis ------------------------------------------------
 ModelSpec = _s.get_message_class("tensorflow.serving.ModelSpec")
 PredictRequest = _s.get_message_class("tensorflow.serving.PredictRequest")

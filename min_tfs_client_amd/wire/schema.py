@@ -787,6 +787,58 @@ FILE_PREDICTION_LOG = proto_file(
     ],
 )
 
+# --- profiler service (reference server.cc:324,339 registers it alongside
+# Model/Prediction; proto tensorflow/core/profiler/profiler_service.proto).
+# Subset closure: ProfileResponse fields 2/4/5 (GraphDef, op_profile,
+# RunMetadata — deep TF-internal closures this server never emits) are
+# omitted; proto3 peers treat absent fields as unset and python-protobuf
+# preserves any unknown fields a TF peer might send. --------------------------
+
+_tool_opts_entry, _tool_opts_field = map_field(
+    "tool_options", 8, "string", ".tensorflow.ToolRequestOptions",
+    ".tensorflow.ProfileRequest")
+
+FILE_PROFILER_SERVICE = proto_file(
+    "tensorflow/core/profiler/profiler_service.proto", "tensorflow",
+    messages=[
+        message("ProfileOptions",
+                fields=[field("include_dataset_ops", 1, "bool")]),
+        message("ToolRequestOptions",
+                fields=[field("output_formats", 2, "string"),
+                        field("save_to_repo", 3, "bool")]),
+        message("ProfileRequest",
+                fields=[field("duration_ms", 1, "uint64"),
+                        field("max_events", 2, "uint64"),
+                        field("tools", 3, "string", repeated=True),
+                        _tool_opts_field,
+                        field("opts", 4, ".tensorflow.ProfileOptions"),
+                        field("repository_root", 5, "string"),
+                        field("session_id", 6, "string"),
+                        field("host_name", 7, "string")],
+                nested=[_tool_opts_entry]),
+        message("ProfileToolData",
+                fields=[field("name", 1, "string"),
+                        field("data", 2, "bytes")]),
+        message("ProfileResponse",
+                fields=[field("encoded_trace", 3, "bytes"),
+                        field("tool_data", 6, ".tensorflow.ProfileToolData",
+                              repeated=True),
+                        field("empty_trace", 7, "bool")]),
+        message("MonitorRequest",
+                fields=[field("duration_ms", 1, "uint64"),
+                        field("monitoring_level", 2, "int32"),
+                        field("timestamp", 3, "bool")]),
+        message("MonitorResponse",
+                fields=[field("data", 1, "string")]),
+    ],
+    services=[service("ProfilerService", [
+        ("Profile", ".tensorflow.ProfileRequest",
+         ".tensorflow.ProfileResponse"),
+        ("Monitor", ".tensorflow.MonitorRequest",
+         ".tensorflow.MonitorResponse"),
+    ])],
+)
+
 # Dependency-ordered registration list.
 _ALL_FILES = [
     FILE_TYPES, FILE_TENSOR_SHAPE, FILE_RESOURCE_HANDLE, FILE_TENSOR,
@@ -797,7 +849,7 @@ _ALL_FILES = [
     FILE_FS_STORAGE_PATH_SOURCE, FILE_MODEL_SERVER_CONFIG,
     FILE_MODEL_MANAGEMENT, FILE_MODEL_SERVICE, FILE_PREDICTION_SERVICE,
     FILE_MONITORING_CONFIG, FILE_CORE_LOGGING, FILE_PREDICTION_LOG,
-    FILE_SSL_CONFIG,
+    FILE_SSL_CONFIG, FILE_PROFILER_SERVICE,
 ]
 
 _pool = descriptor_pool.Default()
