@@ -18,6 +18,7 @@
 #include <pybind11/functional.h>
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
+#include <torch/extension.h>
 
 #include <algorithm>
 #include <atomic>
@@ -219,6 +220,70 @@ class PinnedPool {
 };
 
 // ---------------------------------------------------------------------------
+// progressive device parse (receive-side overlap)
+// ---------------------------------------------------------------------------
+// While a PredictResponse is still streaming in off the socket, the client
+// reader walks the canonical wire layout (model_spec, then outputs map
+// entries whose dtype/shape headers precede their tensor_content) and
+// issues hipMemcpyAsync H2D for each content span's already-received bytes
+// straight out of the pinned receive buffer — by the time the trailers
+// arrive, the output tensors are already (mostly) resident in HBM. Any
+// non-canonical layout (typed *_val fields, unknown fields, repeat-fill
+// shapes) flips `failed` and the caller falls back to the ordinary
+// post-receive parse; correctness never depends on the prospector.
+
+inline bool tf_dtype_to_scalar(int dt, at::ScalarType* st) {
+  switch (dt) {
+    case 1: *st = at::kFloat; return true;
+    case 2: *st = at::kDouble; return true;
+    case 3: *st = at::kInt; return true;
+    case 4: *st = at::kByte; return true;
+    case 5: *st = at::kShort; return true;
+    case 6: *st = at::kChar; return true;
+    case 8: *st = at::kComplexFloat; return true;
+    case 9: *st = at::kLong; return true;
+    case 10: *st = at::kBool; return true;
+    case 14: *st = at::kBFloat16; return true;
+    case 17: *st = at::kUInt16; return true;
+    case 18: *st = at::kComplexDouble; return true;
+    case 19: *st = at::kHalf; return true;
+    case 22: *st = at::kUInt32; return true;
+    case 23: *st = at::kUInt64; return true;
+    default: return false;
+  }
+}
+
+struct MsgAssemblyFwd;  // below
+
+struct DeviceParse {
+  bool enabled = false;
+  bool failed = false;
+  bool done = false;
+  int device = 0;
+  hipStream_t stream = nullptr;  // owned by the channel
+  // structural state
+  size_t pos = 0;       // absolute offset parsed so far
+  int state = 0;        // 0 TOP, 1 ENTRY, 2 TENSOR
+  size_t entry_end = 0, tp_end = 0;
+  std::string cur_name;
+  int cur_dtype = 0;
+  std::vector<int64_t> cur_shape;
+  bool cur_has_tensor = false;
+  at::Tensor cur_tensor;
+  // content copy state
+  bool in_content = false;
+  size_t content_off = 0, content_len = 0, content_copied = 0;
+  std::vector<std::pair<std::string, at::Tensor>> outs;
+
+  static constexpr size_t kCopyBatch = 1u << 20;  // batch tiny arrivals
+
+  void fail() {
+    failed = true;
+    in_content = false;
+  }
+};
+
+// ---------------------------------------------------------------------------
 // in-flight message assembly (shared shape between server request streams
 // and client response streams)
 // ---------------------------------------------------------------------------
@@ -265,6 +330,190 @@ struct MsgAssembly {
 
   bool complete() const { return have_len && msg_have == msg.len; }
 };
+
+// One incremental step of the receive-side device parse. Called by the
+// client reader after every DATA feed; transactional — a structural
+// element that is only partially received is retried on the next feed.
+inline void device_parse_advance(DeviceParse& dp, MsgAssembly& body) {
+  if (!dp.enabled || dp.failed || dp.done) return;
+  if (!body.have_len) return;
+  if (dp.pos == 0 && dp.outs.empty() && !dp.in_content) {
+    // async H2D needs a pinned source; small messages skip the machinery
+    if (body.msg.deleter == nullptr || body.msg.len < (1u << 20)) {
+      dp.enabled = false;
+      return;
+    }
+  }
+  const uint8_t* base = body.msg.p;
+  const size_t watermark = body.msg_have;
+  const size_t total = body.msg.len;
+
+  auto issue_copies = [&]() -> bool {
+    // returns true when the current span finished copying
+    size_t span_end = dp.content_off + dp.content_len;
+    size_t have_end = watermark < span_end ? watermark : span_end;
+    size_t done_abs = dp.content_off + dp.content_copied;
+    if (have_end > done_abs) {
+      size_t n = have_end - done_abs;
+      bool flush = have_end == span_end || body.complete();
+      if (n >= DeviceParse::kCopyBatch || flush) {
+        if (hipSetDevice(dp.device) != hipSuccess) { dp.fail(); return false; }
+        char* dst = static_cast<char*>(dp.cur_tensor.data_ptr())
+                    + dp.content_copied;
+        if (hipMemcpyAsync(dst, base + done_abs, n, hipMemcpyHostToDevice,
+                           dp.stream) != hipSuccess) {
+          dp.fail();
+          return false;
+        }
+        dp.content_copied += n;
+      }
+    }
+    return dp.content_copied == dp.content_len;
+  };
+
+  try {
+    while (true) {
+      if (dp.in_content) {
+        if (!issue_copies()) return;  // need more bytes (or failed)
+        if (dp.failed) return;
+        dp.in_content = false;
+        dp.pos = dp.content_off + dp.content_len;
+        dp.cur_has_tensor = true;
+      }
+      tfswire::Cursor c{base + dp.pos, base + watermark};
+      const uint8_t* step_start = c.p;
+      try {
+        if (dp.state == 0) {  // TOP level of PredictResponse
+          if (dp.pos == total) {
+            dp.done = true;
+            return;
+          }
+          int wt = 0;
+          int f = c.read_tag(&wt);
+          if (f == 2 && wt == tfswire::WT_LEN) {       // model_spec
+            (void)c.read_len_delim();                  // waits if partial
+            dp.pos = size_t(c.p - base);
+          } else if (f == 1 && wt == tfswire::WT_LEN) { // outputs entry
+            uint64_t len = c.read_varint();
+            size_t body_off = size_t(c.p - base);
+            if (body_off + len > total) { dp.fail(); return; }
+            dp.entry_end = body_off + len;
+            dp.pos = body_off;
+            dp.state = 1;
+          } else {
+            dp.fail();
+            return;
+          }
+        } else if (dp.state == 1) {  // entry: key then value(TensorProto)
+          int wt = 0;
+          int f = c.read_tag(&wt);
+          if (f != 1 || wt != tfswire::WT_LEN) { dp.fail(); return; }
+          tfswire::Cursor key = c.read_len_delim();
+          dp.cur_name.assign(reinterpret_cast<const char*>(key.p),
+                             size_t(key.end - key.p));
+          f = c.read_tag(&wt);
+          if (f != 2 || wt != tfswire::WT_LEN) { dp.fail(); return; }
+          uint64_t len = c.read_varint();
+          size_t body_off = size_t(c.p - base);
+          if (body_off + len > dp.entry_end) { dp.fail(); return; }
+          dp.tp_end = body_off + len;
+          dp.pos = body_off;
+          dp.state = 2;
+          dp.cur_dtype = 0;
+          dp.cur_shape.clear();
+          dp.cur_has_tensor = false;
+          dp.cur_tensor = at::Tensor();
+        } else {  // state 2: inside TensorProto
+          if (dp.pos == dp.tp_end) {
+            // finalize this output
+            if (!dp.cur_has_tensor) {
+              int64_t numel = 1;
+              for (int64_t d : dp.cur_shape) numel *= d;
+              at::ScalarType st;
+              if (numel == 0 && tf_dtype_to_scalar(dp.cur_dtype, &st)) {
+                dp.cur_tensor = at::empty(
+                    dp.cur_shape, at::TensorOptions().dtype(st).device(
+                                      at::kCUDA, dp.device));
+                dp.cur_has_tensor = true;
+              } else {
+                dp.fail();
+                return;
+              }
+            }
+            dp.outs.emplace_back(dp.cur_name, dp.cur_tensor);
+            dp.cur_tensor = at::Tensor();
+            if (dp.pos != dp.entry_end) { dp.fail(); return; }
+            dp.state = 0;
+            continue;
+          }
+          int wt = 0;
+          int f = c.read_tag(&wt);
+          if (f == 1 && wt == tfswire::WT_VARINT) {
+            dp.cur_dtype = int(c.read_varint());
+          } else if (f == 2 && wt == tfswire::WT_LEN) {
+            tfswire::Cursor sh = c.read_len_delim();
+            dp.cur_shape.clear();
+            while (!sh.done()) {
+              int swt = 0;
+              int sf = sh.read_tag(&swt);
+              if (sf == 2 && swt == tfswire::WT_LEN) {
+                tfswire::Cursor dim = sh.read_len_delim();
+                int64_t size = 0;
+                while (!dim.done()) {
+                  int dwt = 0;
+                  int df = dim.read_tag(&dwt);
+                  if (df == 1 && dwt == tfswire::WT_VARINT)
+                    size = int64_t(dim.read_varint());
+                  else
+                    dim.skip(dwt);
+                }
+                dp.cur_shape.push_back(size);
+              } else if (sf == 3) {  // unknown_rank: cannot preallocate
+                dp.fail();
+                return;
+              } else {
+                sh.skip(swt);
+              }
+            }
+          } else if (f == 3 && wt == tfswire::WT_VARINT) {
+            (void)c.read_varint();  // version_number
+          } else if (f == 4 && wt == tfswire::WT_LEN) {  // tensor_content
+            uint64_t len = c.read_varint();
+            size_t off = size_t(c.p - base);
+            if (off + len > dp.tp_end) { dp.fail(); return; }
+            at::ScalarType st;
+            if (!tf_dtype_to_scalar(dp.cur_dtype, &st)) { dp.fail(); return; }
+            int64_t numel = 1;
+            for (int64_t d : dp.cur_shape) numel *= d;
+            size_t esize = at::elementSize(st);
+            if (uint64_t(numel) * esize != len) { dp.fail(); return; }
+            dp.cur_tensor = at::empty(
+                dp.cur_shape,
+                at::TensorOptions().dtype(st).device(at::kCUDA, dp.device));
+            dp.content_off = off;
+            dp.content_len = size_t(len);
+            dp.content_copied = 0;
+            dp.in_content = true;
+            dp.pos = off;
+            continue;  // enter the copy loop
+          } else {
+            dp.fail();  // typed *_val or unknown field
+            return;
+          }
+          dp.pos = size_t(c.p - base);
+        }
+      } catch (const std::exception&) {
+        // truncated inside a structural element: if the message is
+        // complete this is malformed, else retry on the next feed
+        (void)step_start;
+        if (body.complete()) dp.fail();
+        return;
+      }
+    }
+  } catch (...) {
+    dp.fail();
+  }
+}
 
 // ---------------------------------------------------------------------------
 // header-block assembly (HEADERS + CONTINUATION)
@@ -1068,11 +1317,38 @@ class GrpcChannel {
   // streaming variant: the request payload is a skeleton plus regions
   // (see write_message_with_regions) — device regions overlap DMA with
   // the send, host regions go out zero-copy straight from tensor memory
+  // lazily created copy stream for receive-side device parse; nullptr
+  // when a different device was already bound (caller falls back)
+  hipStream_t device_stream(int dev) {
+    std::lock_guard<std::mutex> lk(dstream_mu_);
+    if (dstream_ != nullptr) return dstream_dev_ == dev ? dstream_ : nullptr;
+    if (hipSetDevice(dev) != hipSuccess) return nullptr;
+    if (hipStreamCreateWithFlags(&dstream_, hipStreamNonBlocking) !=
+        hipSuccess) {
+      dstream_ = nullptr;
+      return nullptr;
+    }
+    dstream_dev_ = dev;
+    return dstream_;
+  }
+
   uint32_t start_call_streaming(const std::string& path, const uint8_t* buf,
                                 size_t len,
                                 const std::vector<WireRegion>& regions,
-                                double timeout_s) {
+                                double timeout_s, int parse_device = -1) {
     uint32_t id = begin_stream(path, timeout_s);
+    if (parse_device >= 0) {
+      hipStream_t ds = device_stream(parse_device);
+      if (ds != nullptr) {
+        auto p = find_pending(id);
+        if (p) {
+          std::lock_guard<std::mutex> lk(p->m);
+          p->dparse.enabled = true;
+          p->dparse.device = parse_device;
+          p->dparse.stream = ds;
+        }
+      }
+    }
     try {
       write_message_with_regions(*conn_, id, buf, len, regions, true);
     } catch (const std::exception& e) {
@@ -1082,6 +1358,75 @@ class GrpcChannel {
       throw RpcCallError(GRPC_UNAVAILABLE, e.what());
     }
     return id;
+  }
+
+  struct ParsedResult {
+    Buf resp;
+    bool parsed_ok = false;
+    std::vector<std::pair<std::string, at::Tensor>> outs;
+  };
+
+  // wait() variant for device-parse calls: quiesces the copy stream in
+  // EVERY exit path (in-flight H2D references the response buffer) and
+  // hands back the prospected device tensors when the full message
+  // parsed canonically.
+  ParsedResult wait_parsed(uint32_t id, double timeout_s) {
+    std::shared_ptr<Pending> p;
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      auto it = pending_.find(id);
+      if (it == pending_.end())
+        throw RpcCallError(GRPC_INTERNAL, "unknown call handle");
+      p = it->second;
+    }
+    auto quiesce = [&] {
+      std::lock_guard<std::mutex> lk(p->m);
+      if (p->dparse.enabled && p->dparse.stream != nullptr)
+        (void)hipStreamSynchronize(p->dparse.stream);
+    };
+    std::unique_lock<std::mutex> lk(p->m);
+    bool ok = true;
+    if (timeout_s > 0) {
+      ok = p->cv.wait_for(lk, std::chrono::duration<double>(timeout_s),
+                          [&] { return p->done; });
+    } else {
+      p->cv.wait(lk, [&] { return p->done; });
+    }
+    if (!ok) {
+      lk.unlock();
+      quiesce();
+      {
+        std::lock_guard<std::mutex> glk(mu_);
+        pending_.erase(id);
+      }
+      try {
+        conn_->send_rst_stream(id, 8 /* CANCEL */);
+      } catch (...) {
+      }
+      conn_->close_send_stream(id);
+      cv_.notify_all();
+      throw RpcCallError(GRPC_DEADLINE_EXCEEDED, "Deadline Exceeded");
+    }
+    ParsedResult r;
+    r.resp = std::move(p->body.msg);
+    int status = p->grpc_status;
+    std::string msg = p->message;
+    bool parsed = p->dparse.enabled && !p->dparse.failed && p->dparse.done;
+    if (parsed) r.outs = std::move(p->dparse.outs);
+    bool need_sync = p->dparse.enabled && p->dparse.stream != nullptr;
+    hipStream_t ds = p->dparse.stream;
+    lk.unlock();
+    if (need_sync && hipStreamSynchronize(ds) != hipSuccess) parsed = false;
+    {
+      std::lock_guard<std::mutex> glk(mu_);
+      pending_.erase(id);
+    }
+    conn_->close_send_stream(id);
+    cv_.notify_all();
+    if (status != GRPC_OK)
+      throw RpcCallError(status < 0 ? GRPC_INTERNAL : status, msg);
+    r.parsed_ok = parsed;
+    return r;
   }
 
   Buf wait(uint32_t id, double timeout_s) {
@@ -1140,10 +1485,15 @@ class GrpcChannel {
     std::vector<h2::Header> resp_headers;
     HeaderBlock hb;
     MsgAssembly body;
+    DeviceParse dparse;
   };
 
   std::shared_ptr<Conn> conn_;
   std::thread reader_;
+  // receive-side device-parse copy stream (lazy; one per channel)
+  std::mutex dstream_mu_;
+  hipStream_t dstream_ = nullptr;
+  int dstream_dev_ = -1;
   std::string authority_;
   std::atomic<bool> closed_{false};
   std::mutex mu_;
@@ -1237,6 +1587,8 @@ class GrpcChannel {
             } else {
               std::lock_guard<std::mutex> lk(p->m);
               p->body.feed_from_socket(conn_->fd, body_len);
+              if (p->dparse.enabled)
+                device_parse_advance(p->dparse, p->body);
               if (pad) h2::discard(conn_->fd, pad);
             }
             conn_->account_received(fh.length);
@@ -1451,6 +1803,43 @@ class NativeRpcError(Exception):
           "overlap staging DMA with the send; host regions are sent "
           "zero-copy from tensor memory. Caller must keep region memory "
           "alive for the duration of the call.")
+      .def(
+          "call_streaming_parsed",
+          [](GrpcChannel& ch, const std::string& path, py::buffer data,
+             py::list region_list, int parse_device, double timeout) {
+            py::buffer_info info = data.request();
+            const uint8_t* ptr = static_cast<const uint8_t*>(info.ptr);
+            size_t len = size_t(info.size) * size_t(info.itemsize);
+            std::vector<WireRegion> regions;
+            regions.reserve(py::len(region_list));
+            for (auto item : region_list) {
+              auto t = py::reinterpret_borrow<py::tuple>(item);
+              regions.push_back(WireRegion{
+                  t[0].cast<size_t>(), t[1].cast<size_t>(),
+                  uintptr_t(t[2].cast<uint64_t>()), t[3].cast<bool>()});
+            }
+            GrpcChannel::ParsedResult r;
+            {
+              py::gil_scoped_release release;
+              uint32_t id = ch.start_call_streaming(path, ptr, len, regions,
+                                                    timeout, parse_device);
+              r = ch.wait_parsed(id, timeout);
+            }
+            py::object outs = py::none();
+            if (r.parsed_ok) {
+              py::dict d;
+              for (auto& kv : r.outs) d[py::str(kv.first)] = kv.second;
+              outs = std::move(d);
+            }
+            return py::make_tuple(outs, OwnedBuf(std::move(r.resp)));
+          },
+          py::arg("path"), py::arg("data"), py::arg("regions"),
+          py::arg("parse_device"), py::arg("timeout") = 0.0,
+          "call_streaming + receive-side progressive unpack: the reader "
+          "issues async H2D for tensor_content spans while the response "
+          "is still arriving. Returns ({name: device tensor}, raw buf); "
+          "the dict is None when the response was non-canonical — parse "
+          "the raw buffer instead.")
       .def(
           "start_streaming",
           [](GrpcChannel& ch, const std::string& path, py::buffer data,

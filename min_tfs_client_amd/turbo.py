@@ -120,6 +120,15 @@ class _NativeStub:
                                             timeout or 0.0)
 
     @_translate_native_error
+    def call_streaming_parsed(self, blob, regions, parse_device,
+                              timeout=60.0):
+        """Streaming send + receive-side progressive unpack: the reader
+        thread H2Ds tensor_content spans while the response is still
+        arriving. Returns ({name: device tensor} | None, raw_buf)."""
+        return self._channel.call_streaming_parsed(
+            self._path, blob, regions, parse_device, timeout or 0.0)
+
+    @_translate_native_error
     def future_streaming(self, blob, regions, timeout=60.0):
         call_id = self._channel.start_streaming(self._path, blob, regions,
                                                 timeout or 0.0)
@@ -304,9 +313,30 @@ class TurboPredictClient:
                 blob, regions, keepalive = self._serialize_streaming(
                     model_name, inputs, model_version, signature_name)
             self.metrics.observe_bytes("tx", len(blob))
-            with trace_span("turbo.rpc", bytes=len(blob), streaming=True):
-                resp = self._predict.call_streaming(blob, regions, timeout)
-            del keepalive
+            dev = (str(output_device) if output_device is not None
+                   else "cpu")
+            parse_dev = -1
+            if (not zero_copy and dev.startswith("cuda")
+                    and torch is not None):
+                parse_dev = torch.device(dev).index or 0
+            if parse_dev >= 0:
+                # receive-side overlap: output tensors are H2D'd while
+                # the response streams in; None => non-canonical layout,
+                # fall through to the ordinary parse of `resp`
+                with trace_span("turbo.rpc", bytes=len(blob),
+                                streaming=True, parse_device=parse_dev):
+                    outs, resp = self._predict.call_streaming_parsed(
+                        blob, regions, parse_dev, timeout)
+                del keepalive
+                if outs is not None:
+                    self.metrics.observe_bytes("rx", len(resp))
+                    return outs
+            else:
+                with trace_span("turbo.rpc", bytes=len(blob),
+                                streaming=True):
+                    resp = self._predict.call_streaming(blob, regions,
+                                                        timeout)
+                del keepalive
         else:
             with trace_span("turbo.serialize", model=model_name,
                             bytes=sum(t.numel() * t.element_size()

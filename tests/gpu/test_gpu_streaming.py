@@ -111,6 +111,96 @@ def test_end_to_end_gpu_streaming_predict(tmp_path):
             assert torch.equal(out["images"], x * 2)
 
 
+def test_receive_side_parsed_unpack():
+    """call_streaming_parsed: the reader H2Ds tensor_content spans while
+    the response streams in; results must equal the ordinary parse."""
+    outputs = _inputs()
+    payload = _buffered(outputs, is_request=False)  # response layout
+    srv = T.GrpcServer("127.0.0.1:0", 2)
+    srv.register_handler("/t.S/Fixed", lambda v: payload)
+    addr = srv.start()
+    try:
+        req_blob, req_regions, keep = _streaming(
+            {"q": torch.zeros(4, device=DEV)})
+        ch = T.GrpcChannel(addr)
+        try:
+            outs, raw = ch.call_streaming_parsed(
+                "/t.S/Fixed", req_blob, list(req_regions), 0, 60.0)
+            assert outs is not None, "canonical response must parse-ahead"
+            _spec, ref, _f = native.parse_predict_response(
+                bytes(raw), "cuda:0", 1)
+            assert set(outs) == set(ref)
+            for k in ref:
+                assert outs[k].is_cuda
+                assert outs[k].dtype == ref[k].dtype
+                assert torch.equal(outs[k], ref[k]), k
+        finally:
+            ch.close()
+    finally:
+        srv.stop()
+
+
+def test_receive_side_parse_fallback_non_canonical():
+    """A response with typed *_val fields (no tensor_content) must return
+    None from the prospector, and the raw buffer must parse normally."""
+    from min_tfs_client_amd.wire import messages as pb
+
+    resp = pb.PredictResponse()
+    resp.model_spec.name = "m"
+    tp = resp.outputs["x"]
+    tp.dtype = 1
+    tp.tensor_shape.dim.add().size = 3
+    tp.float_val.extend([1.0, 2.0, 3.0])
+    # pad so the message crosses the 1MB pinned threshold and the
+    # prospector actually engages before bailing on float_val
+    big = resp.outputs["pad"]
+    big.dtype = 1
+    big.tensor_shape.dim.add().size = 1 << 19
+    big.tensor_content = b"\x00" * (4 << 19)
+    payload = resp.SerializeToString()
+
+    srv = T.GrpcServer("127.0.0.1:0", 2)
+    srv.register_handler("/t.S/Fixed", lambda v: payload)
+    addr = srv.start()
+    try:
+        inputs = {"x": torch.randn(4, device=DEV)}
+        blob, regions, keepalive = _streaming(inputs)
+        ch = T.GrpcChannel(addr)
+        try:
+            outs, raw = ch.call_streaming_parsed(
+                "/t.S/Fixed", blob, list(regions), 0, 60.0)
+            # typed-field entry comes first -> prospector bails -> None
+            assert outs is None
+            assert bytes(raw) == payload
+        finally:
+            ch.close()
+    finally:
+        srv.stop()
+
+
+def test_predict_uses_parsed_receive(tmp_path):
+    """turbo.predict(output_device=cuda) end-to-end through the parse-
+    ahead path with a multi-tensor response."""
+    sock = f"unix://{tmp_path}/pr.sock"
+    with ModelServer(address=sock, raw_predict=True, device=DEV) as srv:
+        srv.manager.load(
+            "mix", Servable(lambda d: {
+                "a": d["a"] * 2,
+                "b": d["b"].to(torch.float64),
+                "c": torch.empty(0, 4, device=DEV),
+            }), version=1)
+        with TurboPredictClient(sock, backend="native") as c:
+            a = torch.randn(64, 3, 224, 224, device=DEV)
+            b = torch.randint(0, 100, (128, 512), dtype=torch.int32,
+                              device=DEV)
+            out = c.predict("mix", {"a": a, "b": b}, output_device=DEV,
+                            timeout=60)
+            assert torch.equal(out["a"], a * 2)
+            assert torch.equal(out["b"], b.to(torch.float64))
+            assert out["c"].shape == (0, 4)
+            assert all(v.is_cuda for v in out.values())
+
+
 def test_concurrent_device_streaming():
     """8 concurrent streaming sends: each leases its own staging context
     (pooled pipeline) — results must not cross-corrupt."""
