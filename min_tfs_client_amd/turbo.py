@@ -16,6 +16,7 @@ write (ops/csrc/pack_kernels.hip StagingPool).
 """
 from __future__ import annotations
 
+import os
 from typing import Dict, Optional, Union
 
 import grpc
@@ -317,7 +318,8 @@ class TurboPredictClient:
                    else "cpu")
             parse_dev = -1
             if (not zero_copy and dev.startswith("cuda")
-                    and torch is not None):
+                    and torch is not None
+                    and os.environ.get("MI355X_RX_PARSE", "1") != "0"):
                 parse_dev = torch.device(dev).index or 0
             if parse_dev >= 0:
                 # receive-side overlap: output tensors are H2D'd while
