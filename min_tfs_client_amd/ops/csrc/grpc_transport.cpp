@@ -1841,9 +1841,28 @@ class NativeRpcError(Exception):
           "the dict is None when the response was non-canonical — parse "
           "the raw buffer instead.")
       .def(
+          "wait_parsed",
+          [](GrpcChannel& ch, uint32_t id, double timeout) {
+            GrpcChannel::ParsedResult r;
+            {
+              py::gil_scoped_release release;
+              r = ch.wait_parsed(id, timeout);
+            }
+            py::object outs = py::none();
+            if (r.parsed_ok) {
+              py::dict d;
+              for (auto& kv : r.outs) d[py::str(kv.first)] = kv.second;
+              outs = std::move(d);
+            }
+            return py::make_tuple(outs, OwnedBuf(std::move(r.resp)));
+          },
+          py::arg("id"), py::arg("timeout") = 0.0,
+          "wait() for calls started with parse_device >= 0: returns "
+          "({name: device tensor} | None, raw buf).")
+      .def(
           "start_streaming",
           [](GrpcChannel& ch, const std::string& path, py::buffer data,
-             py::list region_list, double timeout) {
+             py::list region_list, double timeout, int parse_device) {
             py::buffer_info info = data.request();
             const uint8_t* ptr = static_cast<const uint8_t*>(info.ptr);
             size_t len = size_t(info.size) * size_t(info.itemsize);
@@ -1857,10 +1876,10 @@ class NativeRpcError(Exception):
             }
             py::gil_scoped_release release;
             return ch.start_call_streaming(path, ptr, len, regions,
-                                           timeout);
+                                           timeout, parse_device);
           },
           py::arg("path"), py::arg("data"), py::arg("regions"),
-          py::arg("timeout") = 0.0,
+          py::arg("timeout") = 0.0, py::arg("parse_device") = -1,
           "Streaming-region variant of start(); the send completes before "
           "this returns (region memory may be released), the response is "
           "collected with wait().")

@@ -96,6 +96,22 @@ class _NativeFuture:
         return False
 
 
+class _NativeParsedFuture:
+    """Future over a device-parse call (see GrpcChannel.wait_parsed)."""
+
+    def __init__(self, channel, call_id: int, timeout: float):
+        self._channel = channel
+        self._id = call_id
+        self._timeout = timeout
+
+    @_translate_native_error
+    def result_parsed(self):
+        return self._channel.wait_parsed(self._id, self._timeout)
+
+    def cancel(self):
+        return False
+
+
 class _NativeStub:
     """grpcio multicallable-shaped adapter over a native GrpcChannel.
     Responses are zero-copy OwnedBuf buffers (the C++ parse and
@@ -134,6 +150,16 @@ class _NativeStub:
         call_id = self._channel.start_streaming(self._path, blob, regions,
                                                 timeout or 0.0)
         return _NativeFuture(self._channel, call_id, timeout or 0.0)
+
+    @_translate_native_error
+    def future_streaming_parsed(self, blob, regions, parse_device,
+                                timeout=60.0):
+        """future_streaming + receive-side progressive unpack; the
+        future's .result_parsed() returns (outs_dict | None, raw_buf)."""
+        call_id = self._channel.start_streaming(self._path, blob, regions,
+                                                timeout or 0.0,
+                                                parse_device)
+        return _NativeParsedFuture(self._channel, call_id, timeout or 0.0)
 
 
 class TurboPredictClient:
@@ -436,6 +462,12 @@ class TurboPredictClient:
                                 for k in keys})
             off += n
 
+        dev = str(output_device) if output_device is not None else "cpu"
+        parse_dev = -1
+        if (use_streaming and dev.startswith("cuda") and torch is not None
+                and os.environ.get("MI355X_RX_PARSE", "1") != "0"):
+            parse_dev = torch.device(dev).index or 0
+
         def send_shard(i):
             # serialize + send one shard. On the native backend the send
             # is synchronous through the socket write, so running the
@@ -450,6 +482,9 @@ class TurboPredictClient:
                 # payload memory through the synchronous send
                 blob, regions, keepalive = self._serialize_streaming(
                     model_name, shard, model_version, signature_name)
+                if parse_dev >= 0:
+                    return stub.future_streaming_parsed(
+                        blob, regions, parse_dev, timeout)
                 return stub.future_streaming(blob, regions, timeout)
             blob = self.serialize_request(model_name, shard, model_version,
                                           signature_name, copy_mode)
@@ -461,13 +496,19 @@ class TurboPredictClient:
             futs = [s.result() for s in sends]
         else:
             futs = [send_shard(i) for i in range(shards)]
-        dev = str(output_device) if output_device is not None else "cpu"
         parts = []
         try:
             for fut in futs:
-                _s, outputs, _ = self._native.parse_predict_response(
-                    fut.result(), dev, copy_mode)
-                parts.append(outputs)
+                if isinstance(fut, _NativeParsedFuture):
+                    outs, raw = fut.result_parsed()
+                    if outs is None:
+                        _s, outs, _ = self._native.parse_predict_response(
+                            raw, dev, copy_mode)
+                    parts.append(outs)
+                else:
+                    _s, outputs, _ = self._native.parse_predict_response(
+                        fut.result(), dev, copy_mode)
+                    parts.append(outputs)
         except Exception:
             for fut in futs:
                 fut.cancel()

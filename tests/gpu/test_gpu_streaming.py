@@ -238,3 +238,20 @@ def test_concurrent_device_streaming():
         assert errs == []
     finally:
         srv.stop()
+
+
+def test_sharded_predict_parse_ahead(tmp_path):
+    """predict_sharded with cuda outputs goes through per-shard
+    parse-ahead futures; results must match the single-shot path."""
+    sock = f"unix://{tmp_path}/shard_pa.sock"
+    with ModelServer(address=sock, raw_predict=True, device=DEV) as srv:
+        srv.manager.load(
+            "double", Servable(lambda d: {k: v * 2 for k, v in d.items()}),
+            version=1)
+        with TurboPredictClient(sock, backend="native",
+                                num_channels=4) as c:
+            x = torch.randn(33, 3, 224, 224, device=DEV)  # uneven shards
+            out = c.predict_sharded("double", {"images": x}, shards=4,
+                                    output_device=DEV, timeout=60)
+            assert out["images"].is_cuda
+            assert torch.equal(out["images"], x * 2)
