@@ -128,16 +128,21 @@ def _raw_predict_bytes_handler(manager, device: str,
     typed-field responses (predict_util.cc:222-226)."""
     from .ops import require_native
 
-    def handler(view):
+    def handler(view, pspec=None, pouts=None):
         t0 = time.perf_counter()
         native = require_native()
         data = memoryview(view)
-        try:
-            spec, inputs, _filter = native.parse_predict_request(
-                data, device, 1)
-        except Exception as e:  # noqa: BLE001
-            raise HandlerAbort(grpc.StatusCode.INVALID_ARGUMENT,
-                               f"request parsing error: {e}")
+        if pouts is not None:
+            # request was prospected while it streamed in: inputs are
+            # already device tensors (C++ reader H2D'd the spans)
+            spec, inputs, _filter = pspec, pouts, pspec["output_filter"]
+        else:
+            try:
+                spec, inputs, _filter = native.parse_predict_request(
+                    data, device, 1)
+            except Exception as e:  # noqa: BLE001
+                raise HandlerAbort(grpc.StatusCode.INVALID_ARGUMENT,
+                                   f"request parsing error: {e}")
         version = spec["version"] if spec["version"] >= 0 else None
         label = spec.get("version_label") or None
         try:
@@ -226,11 +231,18 @@ class NativeTransportServer:
         self.metrics.attach_source(self._cxx_stats)
         self._use_content = output_encoding == "tensor_content"
         ps, ms = prediction_service, model_service
-        self._srv.register_handler(
-            PREDICT_PATH,
-            _raw_predict_bytes_handler(manager, device, self.metrics,
-                                       request_logger,
-                                       use_content=self._use_content))
+        pred_handler = _raw_predict_bytes_handler(
+            manager, device, self.metrics, request_logger,
+            use_content=self._use_content)
+        if device.startswith("cuda"):
+            import torch as _torch
+            dev_idx = _torch.device(device).index or 0
+            # requests stream their tensor_content spans to the GPU while
+            # still arriving (echo-table models are skipped in C++)
+            self._srv.register_handler_parsed(PREDICT_PATH, pred_handler,
+                                              dev_idx)
+        else:
+            self._srv.register_handler(PREDICT_PATH, pred_handler)
         self._srv.register_handler(
             _PS + "Classify",
             _proto_handler(ps.Classify, pb.ClassificationRequest))

@@ -255,12 +255,35 @@ inline bool tf_dtype_to_scalar(int dt, at::ScalarType* st) {
 
 struct MsgAssemblyFwd;  // below
 
+// Shared holder for a prospecting copy stream: the last owner (the
+// connection loop or a still-queued Task) quiesces and destroys it.
+struct CopyStream {
+  hipStream_t s = nullptr;
+  ~CopyStream() {
+    if (s != nullptr) {
+      (void)hipStreamSynchronize(s);
+      (void)hipStreamDestroy(s);
+    }
+  }
+};
+
 struct DeviceParse {
   bool enabled = false;
   bool failed = false;
   bool done = false;
   int device = 0;
-  hipStream_t stream = nullptr;  // owned by the channel
+  hipStream_t stream = nullptr;  // owned by the channel / connection
+  // message layout: response (spec=2, entries=1) by default; requests
+  // (server side) flip to spec=1, entries=2 and may carry output_filter
+  int spec_field = 2;
+  int map_field = 1;
+  // captured model_spec (server-side request prospecting needs it)
+  std::string spec_name, spec_signature, spec_label;
+  int64_t spec_version = -1;
+  std::vector<std::string> out_filter;
+  // server-side: skip prospecting for models served by the C++ echo
+  // fast path (their payloads never touch the GPU)
+  std::function<bool(const std::string&)> skip_model;
   // structural state
   size_t pos = 0;       // absolute offset parsed so far
   int state = 0;        // 0 TOP, 1 ENTRY, 2 TENSOR
@@ -383,23 +406,63 @@ inline void device_parse_advance(DeviceParse& dp, MsgAssembly& body) {
       tfswire::Cursor c{base + dp.pos, base + watermark};
       const uint8_t* step_start = c.p;
       try {
-        if (dp.state == 0) {  // TOP level of PredictResponse
+        if (dp.state == 0) {  // TOP level of the Predict message
           if (dp.pos == total) {
             dp.done = true;
             return;
           }
           int wt = 0;
           int f = c.read_tag(&wt);
-          if (f == 2 && wt == tfswire::WT_LEN) {       // model_spec
-            (void)c.read_len_delim();                  // waits if partial
+          if (f == dp.spec_field && wt == tfswire::WT_LEN) {  // model_spec
+            tfswire::Cursor spec = c.read_len_delim();  // waits if partial
+            while (!spec.done()) {
+              int swt = 0;
+              int sf = spec.read_tag(&swt);
+              if (sf == 1 && swt == tfswire::WT_LEN) {
+                tfswire::Cursor v = spec.read_len_delim();
+                dp.spec_name.assign(reinterpret_cast<const char*>(v.p),
+                                    size_t(v.end - v.p));
+              } else if (sf == 2 && swt == tfswire::WT_LEN) {
+                tfswire::Cursor v = spec.read_len_delim();  // Int64Value
+                while (!v.done()) {
+                  int vwt = 0;
+                  int vf = v.read_tag(&vwt);
+                  if (vf == 1 && vwt == tfswire::WT_VARINT)
+                    dp.spec_version = int64_t(v.read_varint());
+                  else
+                    v.skip(vwt);
+                }
+              } else if (sf == 3 && swt == tfswire::WT_LEN) {
+                tfswire::Cursor v = spec.read_len_delim();
+                dp.spec_signature.assign(
+                    reinterpret_cast<const char*>(v.p),
+                    size_t(v.end - v.p));
+              } else if (sf == 4 && swt == tfswire::WT_LEN) {
+                tfswire::Cursor v = spec.read_len_delim();
+                dp.spec_label.assign(reinterpret_cast<const char*>(v.p),
+                                     size_t(v.end - v.p));
+              } else {
+                spec.skip(swt);
+              }
+            }
+            if (dp.skip_model && dp.skip_model(dp.spec_name)) {
+              dp.enabled = false;  // echo fast path: stay off the GPU
+              return;
+            }
             dp.pos = size_t(c.p - base);
-          } else if (f == 1 && wt == tfswire::WT_LEN) { // outputs entry
+          } else if (f == dp.map_field && wt == tfswire::WT_LEN) {
             uint64_t len = c.read_varint();
             size_t body_off = size_t(c.p - base);
             if (body_off + len > total) { dp.fail(); return; }
             dp.entry_end = body_off + len;
             dp.pos = body_off;
             dp.state = 1;
+          } else if (dp.spec_field == 1 && f == 3 &&
+                     wt == tfswire::WT_LEN) {  // request output_filter
+            tfswire::Cursor v = c.read_len_delim();
+            dp.out_filter.emplace_back(
+                reinterpret_cast<const char*>(v.p), size_t(v.end - v.p));
+            dp.pos = size_t(c.p - base);
           } else {
             dp.fail();
             return;
@@ -709,6 +772,16 @@ class GrpcServer {
     py_handlers_[path] = std::move(fn);
   }
 
+  // handler called as fn(view, spec_dict_or_None, outs_dict_or_None):
+  // requests on `path` are prospected while they stream in and their
+  // tensor_content spans H2D'd to `device` (see DeviceParse)
+  void register_handler_parsed(const std::string& path, py::object fn,
+                               int device) {
+    std::lock_guard<std::mutex> lk(handler_mu_);
+    py_handlers_[path] = std::move(fn);
+    parsed_paths_[path] = device;
+  }
+
   void set_echo_models(
       const std::string& path,
       const std::map<std::string, std::set<int64_t>>& models) {
@@ -777,6 +850,7 @@ class GrpcServer {
 
   std::mutex handler_mu_;
   std::unordered_map<std::string, py::object> py_handlers_;
+  std::unordered_map<std::string, int> parsed_paths_;
   std::unordered_map<std::string, std::map<std::string, std::set<int64_t>>>
       echo_models_;
 
@@ -785,6 +859,8 @@ class GrpcServer {
     uint32_t stream = 0;
     std::string path;
     Buf msg;
+    DeviceParse dparse;
+    std::shared_ptr<CopyStream> stream_ref;
   };
   std::mutex q_mu_;
   std::condition_variable q_cv_;
@@ -872,7 +948,11 @@ class GrpcServer {
       HeaderBlock hb;
       bool headers_done = false;
       MsgAssembly body;
+      DeviceParse dparse;
+      std::shared_ptr<CopyStream> stream_ref;
     };
+    // lazy per-connection copy stream for request prospecting
+    std::shared_ptr<CopyStream> conn_stream;
     std::unordered_map<uint32_t, SrvStream> streams;
     h2::HpackDecoder decoder;
     uint32_t continuation_stream = 0;
@@ -901,6 +981,8 @@ class GrpcServer {
               h2::discard(conn->fd, len);
             } else {
               it->second.body.feed_from_socket(conn->fd, body_len);
+              if (it->second.dparse.enabled)
+                device_parse_advance(it->second.dparse, it->second.body);
               if (pad) h2::discard(conn->fd, pad);
             }
             conn->account_received(fh.length);
@@ -915,7 +997,7 @@ class GrpcServer {
             read_headers_fragment(conn->fd, fh, &st.hb.block);
             st.hb.end_stream = (fh.flags & h2::FL_END_STREAM) != 0;
             if (fh.flags & h2::FL_END_HEADERS) {
-              finish_headers(conn, fh.stream, st, decoder);
+              finish_headers(conn, fh.stream, st, decoder, &conn_stream);
               if (st.hb.end_stream) {
                 dispatch(conn, fh.stream, st);
                 streams.erase(fh.stream);
@@ -935,7 +1017,7 @@ class GrpcServer {
                           reinterpret_cast<uint8_t*>(&st.hb.block[off]),
                           fh.length);
             if (fh.flags & h2::FL_END_HEADERS) {
-              finish_headers(conn, continuation_stream, st, decoder);
+              finish_headers(conn, continuation_stream, st, decoder, &conn_stream);
               if (st.hb.end_stream) {
                 dispatch(conn, continuation_stream, st);
                 streams.erase(continuation_stream);
@@ -994,7 +1076,8 @@ class GrpcServer {
 
   template <typename SrvStreamT>
   void finish_headers(const std::shared_ptr<Conn>& conn, uint32_t stream,
-                      SrvStreamT& st, h2::HpackDecoder& decoder) {
+                      SrvStreamT& st, h2::HpackDecoder& decoder,
+                      std::shared_ptr<CopyStream>* conn_stream) {
     auto headers = decoder.decode(
         reinterpret_cast<const uint8_t*>(st.hb.block.data()),
         st.hb.block.size());
@@ -1004,6 +1087,35 @@ class GrpcServer {
     const std::string* path = find_header(headers, ":path");
     st.path = path ? *path : "";
     conn->open_send_stream(stream);
+    int device = -1;
+    {
+      std::lock_guard<std::mutex> lk(handler_mu_);
+      auto it = parsed_paths_.find(st.path);
+      if (it != parsed_paths_.end()) device = it->second;
+    }
+    if (device >= 0) {
+      if (*conn_stream == nullptr) {
+        auto holder = std::make_shared<CopyStream>();
+        if (hipSetDevice(device) == hipSuccess &&
+            hipStreamCreateWithFlags(&holder->s, hipStreamNonBlocking) ==
+                hipSuccess)
+          *conn_stream = std::move(holder);
+      }
+      if (*conn_stream != nullptr) {
+        st.stream_ref = *conn_stream;
+        st.dparse.enabled = true;
+        st.dparse.device = device;
+        st.dparse.stream = (*conn_stream)->s;
+        st.dparse.spec_field = 1;  // request layout
+        st.dparse.map_field = 2;
+        st.dparse.skip_model = [this](const std::string& name) {
+          std::lock_guard<std::mutex> lk(handler_mu_);
+          for (auto& kv : echo_models_)
+            if (kv.second.count(name)) return true;
+          return false;
+        };
+      }
+    }
   }
 
   template <typename SrvStreamT>
@@ -1014,6 +1126,8 @@ class GrpcServer {
     t.stream = stream;
     t.path = std::move(st.path);
     t.msg = std::move(st.body.msg);
+    t.dparse = std::move(st.dparse);
+    t.stream_ref = std::move(st.stream_ref);
     if (!st.body.have_len) {
       t.msg.alloc(0);  // empty request message (e.g. empty proto)
       t.msg.len = 0;
@@ -1098,6 +1212,16 @@ class GrpcServer {
       }
       fn = it->second;
     }
+    bool parsed_path;
+    {
+      std::lock_guard<std::mutex> lk(handler_mu_);
+      parsed_path = parsed_paths_.count(t.path) != 0;
+    }
+    bool have_parse = parsed_path && t.dparse.enabled &&
+                      !t.dparse.failed && t.dparse.done;
+    if (have_parse && t.dparse.stream != nullptr &&
+        hipStreamSynchronize(t.dparse.stream) != hipSuccess)
+      have_parse = false;
     int err_code = 0;
     std::string err_msg;
     py::object result;
@@ -1106,7 +1230,30 @@ class GrpcServer {
       try {
         py::memoryview view = py::memoryview::from_memory(
             t.msg.p, py::ssize_t(t.msg.len));
-        result = fn(view);
+        if (parsed_path) {
+          py::object spec = py::none();
+          py::object outs = py::none();
+          if (have_parse) {
+            py::dict sd;
+            sd["name"] = t.dparse.spec_name;
+            sd["version"] = t.dparse.spec_version;
+            sd["signature_name"] = t.dparse.spec_signature;
+            sd["version_label"] = t.dparse.spec_label;
+            py::list filt;
+            for (auto& f2 : t.dparse.out_filter) filt.append(py::str(f2));
+            sd["output_filter"] = filt;
+            py::dict od;
+            for (auto& kv : t.dparse.outs)
+              od[py::str(kv.first)] = kv.second;
+            spec = std::move(sd);
+            outs = std::move(od);
+          }
+          result = fn(view, spec, outs);
+        } else {
+          result = fn(view);
+        }
+        // drop tensor refs before the send (outs were copied into python)
+        t.dparse.outs.clear();
       } catch (py::error_already_set& e) {
         err_code = GRPC_UNKNOWN;
         try {
@@ -1746,6 +1893,12 @@ class NativeRpcError(Exception):
            py::call_guard<py::gil_scoped_release>())
       .def("register_handler", &GrpcServer::register_handler,
            py::arg("path"), py::arg("fn"))
+      .def("register_handler_parsed", &GrpcServer::register_handler_parsed,
+           py::arg("path"), py::arg("fn"), py::arg("device"),
+           "Like register_handler, but fn is called as fn(view, spec, "
+           "outs): requests are prospected while streaming in and their "
+           "tensor_content spans H2D'd to `device`; spec/outs are None "
+           "when the request was non-canonical.")
       .def("add_address", &GrpcServer::add_address, py::arg("address"),
            "Add an extra listen address (call before start()).")
       .def("set_echo_models", &GrpcServer::set_echo_models, py::arg("path"),

@@ -255,3 +255,80 @@ def test_sharded_predict_parse_ahead(tmp_path):
                                     output_device=DEV, timeout=60)
             assert out["images"].is_cuda
             assert torch.equal(out["images"], x * 2)
+
+
+def test_server_request_prospecting():
+    """register_handler_parsed: the server H2Ds request tensor_content
+    while it streams in and hands the handler device tensors + spec."""
+    seen = {}
+
+    def handler(view, spec, outs):
+        seen["spec"] = spec
+        seen["outs"] = outs
+        return b"ok"
+
+    srv = T.GrpcServer("127.0.0.1:0", 2)
+    srv.register_handler_parsed(
+        "/tensorflow.serving.PredictionService/Predict", handler, 0)
+    addr = srv.start()
+    try:
+        inputs = _inputs()
+        names = list(inputs.keys())
+        blob, regions, keep = native.serialize_predict_streaming(
+            True, "promodel", 7, "sig", names, [inputs[k] for k in names])
+        torch.cuda.current_stream().synchronize()
+        ch = T.GrpcChannel(addr)
+        try:
+            assert bytes(ch.call_streaming(
+                "/tensorflow.serving.PredictionService/Predict",
+                blob, list(regions), 60.0)) == b"ok"
+        finally:
+            ch.close()
+        assert seen["outs"] is not None, "canonical request must prospect"
+        assert seen["spec"]["name"] == "promodel"
+        assert seen["spec"]["version"] == 7
+        assert seen["spec"]["signature_name"] == "sig"
+        for k in inputs:
+            assert seen["outs"][k].is_cuda
+            assert torch.equal(seen["outs"][k], inputs[k]), k
+    finally:
+        srv.stop()
+
+
+def test_server_prospecting_end_to_end_gpu_servable(tmp_path):
+    """ModelServer(device=cuda): full round trip where BOTH directions
+    are prospected (request spans H2D'd during receive, response spans
+    H2D'd client-side during receive)."""
+    sock = f"unix://{tmp_path}/prospect.sock"
+    with ModelServer(address=sock, raw_predict=True, device=DEV) as srv:
+        srv.manager.load(
+            "triple", Servable(lambda d: {k: v * 3 for k, v in d.items()}),
+            version=1)
+        with TurboPredictClient(sock, backend="native") as c:
+            x = torch.randn(64, 3, 224, 224, device=DEV)
+            out = c.predict("triple", {"images": x}, output_device=DEV,
+                            timeout=60)
+            assert torch.equal(out["images"], x * 3)
+
+
+def test_echo_model_skips_prospecting(tmp_path):
+    """Identity models are served by the all-C++ echo path; the skip_model
+    hook must keep their requests off the GPU while still echoing
+    correctly on a cuda-device server."""
+    from min_tfs_client_amd.server import identity_servable
+
+    sock = f"unix://{tmp_path}/echo_skip.sock"
+    with ModelServer(address=sock, raw_predict=True, device=DEV) as srv:
+        srv.manager.load("default", identity_servable(), version=1)
+        with TurboPredictClient(sock, backend="native") as c:
+            x = torch.randn(32, 3, 224, 224, device=DEV)
+            before = torch.cuda.memory_allocated()
+            for _ in range(3):
+                out = c.predict("default", {"images": x},
+                                output_device=DEV, timeout=60)
+                assert torch.equal(out["images"], x)
+            # server-side prospecting for the echo model would have
+            # allocated ~19MB x 3 on the device inside the SERVER (same
+            # process here): allow client-side parse-ahead allocs only
+            after = torch.cuda.memory_allocated()
+            assert after - before < 200 << 20
