@@ -234,7 +234,9 @@ class NativeTransportServer:
         pred_handler = _raw_predict_bytes_handler(
             manager, device, self.metrics, request_logger,
             use_content=self._use_content)
-        if device.startswith("cuda"):
+        import os as _os
+        if (device.startswith("cuda")
+                and _os.environ.get("MI355X_RX_PARSE", "1") != "0"):
             import torch as _torch
             dev_idx = _torch.device(device).index or 0
             # requests stream their tensor_content spans to the GPU while
