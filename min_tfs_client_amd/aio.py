@@ -25,11 +25,32 @@ from .utils.allocator import tune_malloc
 
 
 class AsyncTurboPredictClient:
+    """``backend``: "native" (default when available, no TLS) runs the
+    C++ transport's streaming send + receive-side parse-ahead off the
+    event loop via the default executor — the C++ side releases the GIL
+    for the whole call; "grpcio" uses grpc.aio (required for TLS)."""
+
     def __init__(self, target: str,
                  credentials: Optional[grpc.ChannelCredentials] = None,
-                 options: Optional[list] = None):
+                 options: Optional[list] = None,
+                 backend: str = "auto"):
         self._native = require_native()
         tune_malloc()
+        if backend == "auto":
+            backend = "grpcio" if credentials is not None else "native"
+            if backend == "native":
+                try:
+                    from . import _transport  # noqa: F401
+                except Exception:
+                    backend = "grpcio"
+        self.backend = backend
+        self._sync = None
+        if backend == "native":
+            from .turbo import TurboPredictClient
+            self._sync = TurboPredictClient(target, backend="native")
+            self._channel = None
+            self._predict = None
+            return
         opts = _CHANNEL_OPTS + (options or [])
         if credentials:
             self._channel = grpc.aio.secure_channel(target, credentials,
@@ -41,6 +62,9 @@ class AsyncTurboPredictClient:
             response_deserializer=_identity)
 
     async def close(self):
+        if self._sync is not None:
+            self._sync.close()
+            return
         await self._channel.close()
 
     async def __aenter__(self):
@@ -59,6 +83,13 @@ class AsyncTurboPredictClient:
         names = list(inputs.keys())
         tensors = [inputs[k] for k in names]
         loop = asyncio.get_running_loop()
+        if self._sync is not None:
+            # one executor hop for the whole call: the native client
+            # releases the GIL through serialize/send/wait/parse
+            return await loop.run_in_executor(
+                None, lambda: self._sync.predict(
+                    model_name, inputs, timeout, model_version,
+                    signature_name, output_device, copy_mode))
         # serialize in the default executor: the C++ side releases the GIL
         # during copies but the call itself can take ~1 ms at 19 MB
         blob = await loop.run_in_executor(

@@ -170,6 +170,22 @@ def test_async_client(raw_server):
     asyncio.run(run())
 
 
+def test_async_client_both_backends(raw_server):
+    import asyncio
+    from min_tfs_client_amd.aio import AsyncTurboPredictClient
+
+    async def run(backend):
+        async with AsyncTurboPredictClient(raw_server.address,
+                                           backend=backend) as c:
+            assert c.backend == backend
+            x = torch.randn(3, 5)
+            out = await c.predict("default", {"x": x})
+            assert torch.equal(out["x"], x)
+
+    asyncio.run(run("native"))
+    asyncio.run(run("grpcio"))
+
+
 def test_deadline_exceeded():
     """gRPC deadline semantics: timeout is seconds, slow servables abort
     with DEADLINE_EXCEEDED (reference requests.py:49 passes timeout
