@@ -296,6 +296,7 @@ struct DeviceParse {
   // content copy state
   bool in_content = false;
   size_t content_off = 0, content_len = 0, content_copied = 0;
+  bool copies_issued = false;
   std::vector<std::pair<std::string, at::Tensor>> outs;
 
   static constexpr size_t kCopyBatch = 1u << 20;  // batch tiny arrivals
@@ -304,6 +305,58 @@ struct DeviceParse {
     failed = true;
     in_content = false;
   }
+
+  // RAII: every destruction path (stream reset, connection death, failed
+  // prospect, task teardown) must drain in-flight H2D copies BEFORE the
+  // source buffer and the destination tensors are freed — a pinned
+  // receive buffer hipHostFree'd with a pending SDMA read (or a tensor
+  // freed with a pending write) is a crash/corruption. The dtor body
+  // runs before member (outs) destruction, and holders declare the
+  // stream's keepalive BEFORE the DeviceParse member so the stream
+  // outlives this sync.
+  void quiesce() {
+    if (copies_issued && stream != nullptr) {
+      (void)hipStreamSynchronize(stream);
+      copies_issued = false;
+    }
+  }
+
+  DeviceParse() = default;
+  DeviceParse(const DeviceParse&) = delete;
+  DeviceParse& operator=(const DeviceParse&) = delete;
+  DeviceParse(DeviceParse&& o) noexcept { *this = std::move(o); }
+  DeviceParse& operator=(DeviceParse&& o) noexcept {
+    if (this != &o) {
+      quiesce();
+      enabled = o.enabled; failed = o.failed; done = o.done;
+      device = o.device; stream = o.stream;
+      spec_field = o.spec_field; map_field = o.map_field;
+      spec_name = std::move(o.spec_name);
+      spec_signature = std::move(o.spec_signature);
+      spec_label = std::move(o.spec_label);
+      spec_version = o.spec_version;
+      out_filter = std::move(o.out_filter);
+      skip_model = std::move(o.skip_model);
+      pos = o.pos; state = o.state;
+      entry_end = o.entry_end; tp_end = o.tp_end;
+      cur_name = std::move(o.cur_name);
+      cur_dtype = o.cur_dtype;
+      cur_shape = std::move(o.cur_shape);
+      cur_has_tensor = o.cur_has_tensor;
+      cur_tensor = std::move(o.cur_tensor);
+      in_content = o.in_content;
+      content_off = o.content_off;
+      content_len = o.content_len;
+      content_copied = o.content_copied;
+      copies_issued = o.copies_issued;
+      outs = std::move(o.outs);
+      o.copies_issued = false;
+      o.enabled = false;
+      o.stream = nullptr;
+    }
+    return *this;
+  }
+  ~DeviceParse() { quiesce(); }
 };
 
 // ---------------------------------------------------------------------------
@@ -388,6 +441,7 @@ inline void device_parse_advance(DeviceParse& dp, MsgAssembly& body) {
           dp.fail();
           return false;
         }
+        dp.copies_issued = true;
         dp.content_copied += n;
       }
     }
@@ -855,12 +909,12 @@ class GrpcServer {
       echo_models_;
 
   struct Task {
+    std::shared_ptr<CopyStream> stream_ref;  // must outlive dparse+msg
     std::shared_ptr<Conn> conn;
     uint32_t stream = 0;
     std::string path;
     Buf msg;
-    DeviceParse dparse;
-    std::shared_ptr<CopyStream> stream_ref;
+    DeviceParse dparse;  // dtor quiesces before msg's buffer is freed
   };
   std::mutex q_mu_;
   std::condition_variable q_cv_;
@@ -944,12 +998,12 @@ class GrpcServer {
 
   void connection_loop(std::shared_ptr<Conn> conn) {
     struct SrvStream {
+      std::shared_ptr<CopyStream> stream_ref;  // must outlive dparse+body
       std::string path;
       HeaderBlock hb;
       bool headers_done = false;
       MsgAssembly body;
-      DeviceParse dparse;
-      std::shared_ptr<CopyStream> stream_ref;
+      DeviceParse dparse;  // dtor quiesces before body's buffer is freed
     };
     // lazy per-connection copy stream for request prospecting
     std::shared_ptr<CopyStream> conn_stream;
@@ -1219,9 +1273,7 @@ class GrpcServer {
     }
     bool have_parse = parsed_path && t.dparse.enabled &&
                       !t.dparse.failed && t.dparse.done;
-    if (have_parse && t.dparse.stream != nullptr &&
-        hipStreamSynchronize(t.dparse.stream) != hipSuccess)
-      have_parse = false;
+    t.dparse.quiesce();  // drain copies even for failed/partial prospects
     int err_code = 0;
     std::string err_msg;
     py::object result;
