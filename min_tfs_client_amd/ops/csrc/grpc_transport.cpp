@@ -127,6 +127,7 @@ constexpr int GRPC_CANCELLED = 1;
 constexpr int GRPC_UNKNOWN = 2;
 constexpr int GRPC_INVALID_ARGUMENT = 3;
 constexpr int GRPC_DEADLINE_EXCEEDED = 4;
+constexpr int GRPC_RESOURCE_EXHAUSTED = 8;
 constexpr int GRPC_UNIMPLEMENTED = 12;
 constexpr int GRPC_INTERNAL = 13;
 constexpr int GRPC_UNAVAILABLE = 14;
@@ -384,6 +385,13 @@ struct MsgAssembly {
             throw RpcCallError(GRPC_UNIMPLEMENTED,
                                "compressed gRPC messages not supported");
           uint32_t len = h2::be32(prefix + 1);
+          // match the grpcio servers' max_receive_message_length (1GB):
+          // an attacker-declared length must not drive the allocation
+          if (len > (1u << 30))
+            throw RpcCallError(
+                GRPC_RESOURCE_EXHAUSTED,
+                "received message larger than max (" +
+                    std::to_string(len) + " vs 1073741824)");
           size_t cap = 0;
           auto& pool = PinnedPool::instance();
           uint8_t* pinned = pool.get(len, &cap);

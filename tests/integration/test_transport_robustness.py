@@ -164,3 +164,19 @@ def test_many_connections_churn(server):
             s.sendall(PREFACE)
         s.close()
     _control_request_ok(addr)
+
+
+def test_huge_declared_message_rejected(server):
+    """A 5-byte gRPC prefix declaring a 2GB message must be refused
+    (RESOURCE_EXHAUSTED semantics) without allocating the 2GB."""
+    host, port, addr = server
+    s = _raw(host, port)
+    # handshake enough for the server to accept a stream: preface,
+    # SETTINGS, empty HEADERS (END_HEADERS), then DATA with huge prefix
+    prefix = b"\x00" + struct.pack("!I", (2 << 30) - 1)
+    s.sendall(PREFACE + _frame(4, 0, 0, b"")
+              + _frame(1, 0x4, 1, b"")
+              + _frame(0, 0, 1, prefix + b"xx"))
+    time.sleep(0.2)
+    s.close()
+    _control_request_ok(addr)
