@@ -9,10 +9,15 @@ small requests, slow for 19 MB image batches. The turbo path:
   serialize--> client --C++ parse--> torch tensors (CPU or device)
 
 This is the MI355X-native realization of the reference's zero-copy encode
-(grpc_tensor_coding.cc:140-248): where TF shares the tensor's backing store
-as a second gRPC slice, we DMA the device tensor directly into the wire
-buffer through pinned staging, overlapping DMA chunks with the host-side
-write (ops/csrc/pack_kernels.hip StagingPool).
+(grpc_tensor_coding.cc:140-248), completed in both directions: sends
+materialize only the wire SKELETON and stream device payloads through
+pooled pinned staging straight into HTTP/2 DATA frames (the
+hipMemcpyAsync of chunk i+1 overlaps the socket write of chunk i —
+measured in profiles/overlap_trace_r02.jsonl), host payloads ride iovec
+from tensor memory with zero user-space copies; receives land in pooled
+pinned buffers and are PARSE-AHEAD unpacked — tensor_content spans H2D
+while the response is still arriving (ops/csrc/staging.h,
+grpc_transport.cpp DeviceParse).
 """
 from __future__ import annotations
 
