@@ -184,3 +184,58 @@ def test_streaming_concurrent_calls():
         assert errs == []
     finally:
         srv.stop()
+
+
+def test_call_streaming_parsed_no_gpu_fallback():
+    """On a machine without a GPU, parse_device requests must degrade to
+    (None, raw_buf) — never error — so callers can always pass a device
+    and fall back to the ordinary parse."""
+    import torch
+    if torch.cuda.is_available():
+        pytest.skip("covered by tests/gpu on GPU machines")
+    srv = T.GrpcServer("127.0.0.1:0", 2)
+    srv.register_handler("/t.S/Echo", lambda v: bytes(v))
+    addr = srv.start()
+    try:
+        inputs = {"x": torch.randn(512, 512)}  # 1MB+, host region
+        blob, regions, keep = _streaming_parts(inputs)
+        ch = T.GrpcChannel(addr)
+        try:
+            outs, raw = ch.call_streaming_parsed(
+                "/t.S/Echo", blob, list(regions), 0, 30.0)
+            assert outs is None
+            assert bytes(raw) == _buffered(inputs)
+        finally:
+            ch.close()
+    finally:
+        srv.stop()
+
+
+def test_register_handler_parsed_no_gpu_fallback():
+    """Server-side: a parsed-registered handler on a GPU-less machine
+    still gets called (with None, None) and serves correctly."""
+    import torch
+    if torch.cuda.is_available():
+        pytest.skip("covered by tests/gpu on GPU machines")
+    calls = []
+
+    def handler(view, spec, outs):
+        calls.append((spec, outs))
+        return bytes(view)
+
+    srv = T.GrpcServer("127.0.0.1:0", 2)
+    srv.register_handler_parsed("/t.S/Echo", handler, 0)
+    addr = srv.start()
+    try:
+        inputs = {"x": torch.randn(512, 512)}
+        blob, regions, keep = _streaming_parts(inputs)
+        ch = T.GrpcChannel(addr)
+        try:
+            got = bytes(ch.call_streaming("/t.S/Echo", blob,
+                                          list(regions), 30.0))
+            assert got == _buffered(inputs)
+        finally:
+            ch.close()
+        assert calls and calls[0] == (None, None)
+    finally:
+        srv.stop()
