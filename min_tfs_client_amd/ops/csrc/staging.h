@@ -47,10 +47,20 @@ namespace mi355x_staging {
                                hipGetErrorString(_e) + " at " #expr);    \
   } while (0)
 
-// 4 MiB chunks x 2 buffers: deep enough to hide DMA behind the consumer
-// (socket write / wire-buffer memcpy), small enough that the first byte
-// reaches the consumer ~80 us after the transfer starts at HBM3E rates.
-constexpr size_t kChunk = 4u << 20;
+// 4 MiB chunks x 2 buffers by default: deep enough to hide DMA behind
+// the consumer (socket write / wire-buffer memcpy), small enough that
+// the first byte reaches the consumer ~80 us after the transfer starts.
+// Tunable via MI355X_STAGING_CHUNK (bytes, clamped to [1 MiB, 32 MiB],
+// read once at first use) for A/B sweeps.
+inline size_t chunk_size() {
+  static const size_t v = [] {
+    const char* e = std::getenv("MI355X_STAGING_CHUNK");
+    long long n = e ? std::atoll(e) : 0;
+    if (n < (1 << 20) || n > (32 << 20)) n = 4 << 20;
+    return size_t(n);
+  }();
+  return v;
+}
 constexpr int kMaxCtx = 8;
 
 struct Ctx {
@@ -63,7 +73,7 @@ struct Ctx {
     MI355X_STAGING_CHECK(
         hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
     for (int i = 0; i < 2; ++i) {
-      MI355X_STAGING_CHECK(hipHostMalloc(&buf[i], kChunk));
+      MI355X_STAGING_CHECK(hipHostMalloc(&buf[i], chunk_size()));
       MI355X_STAGING_CHECK(
           hipEventCreateWithFlags(&evt[i], hipEventDisableTiming));
     }
@@ -91,6 +101,7 @@ struct Ctx {
   // time (~70 us at 4 MiB) on every chunk.
   template <typename Consume>
   void d2h(const void* src_dev, size_t nbytes, Consume&& consume) {
+    const size_t kChunk = chunk_size();
     size_t nchunks = (nbytes + kChunk - 1) / kChunk;
     const char* trace_path = std::getenv("MI355X_STAGING_TRACE");
     std::vector<double> wait_ms, consume_ms;
@@ -151,6 +162,7 @@ struct Ctx {
   // production of the next chunk. Returns after the last DMA completes.
   template <typename Produce>
   void h2d(void* dst_dev, size_t nbytes, Produce&& produce) {
+    const size_t kChunk = chunk_size();
     size_t nchunks = (nbytes + kChunk - 1) / kChunk;
     for (size_t c = 0; c < nchunks; ++c) {
       size_t off = c * kChunk;
