@@ -39,8 +39,39 @@ class TensorServingClient:
     def __init__(self, host: str, port: int,
                  credentials: Optional[grpc.ChannelCredentials] = None,
                  options: Optional[list] = None,
-                 enable_retries: bool = False) -> None:
+                 enable_retries: bool = False,
+                 backend: str = "auto") -> None:
+        """``backend``: "auto" rides the C++ HTTP/2 transport when
+        possible (insecure, no custom channel options, no grpcio retry
+        policy — the features below are grpcio-specific) and falls back
+        to grpcio otherwise; "grpcio"/"native" force a stack."""
         self._host_address = f"{host}:{port}"
+        use_native = (backend in ("auto", "native") and credentials is None
+                      and not options and not enable_retries)
+        if use_native:
+            try:
+                from . import _transport
+            except Exception:
+                if backend == "native":
+                    raise
+                use_native = False
+        if backend == "native" and not use_native:
+            raise ValueError(
+                "backend='native' is incompatible with credentials/"
+                "options/enable_retries (grpcio-specific features)")
+        if use_native:
+            from . import _transport
+            from .turbo import (
+                NativeModelServiceStub,
+                NativePredictionServiceStub,
+            )
+            self._channel = _transport.GrpcChannel(self._host_address)
+            self._prediction_stub = NativePredictionServiceStub(
+                self._channel)
+            self._model_stub = NativeModelServiceStub(self._channel)
+            self.backend = "native"
+            return
+        self.backend = "grpcio"
         default_options = [
             ("grpc.max_send_message_length", 1 << 30),
             ("grpc.max_receive_message_length", 1 << 30),

@@ -167,6 +167,53 @@ class _NativeStub:
         return _NativeParsedFuture(self._channel, call_id, timeout or 0.0)
 
 
+class _NativeProtoMethod:
+    """Protobuf-message unary call over the native channel (used by the
+    reference-parity TensorServingClient when it rides the C++
+    transport)."""
+
+    def __init__(self, channel, path, resp_cls):
+        self._channel = channel
+        self._path = path
+        self._resp_cls = resp_cls
+
+    @_translate_native_error
+    def __call__(self, request, timeout=None):
+        blob = self._channel.call(self._path, request.SerializeToString(),
+                                  timeout or 0.0)
+        return self._resp_cls.FromString(bytes(blob))
+
+
+class NativePredictionServiceStub:
+    """PredictionServiceStub-shaped adapter over a native GrpcChannel
+    (same 5 rpcs as wire/grpc_stubs.PredictionServiceStub)."""
+
+    def __init__(self, channel):
+        from .wire import messages as pb
+        _PS = "/tensorflow.serving.PredictionService/"
+        self.Predict = _NativeProtoMethod(
+            channel, _PS + "Predict", pb.PredictResponse)
+        self.Classify = _NativeProtoMethod(
+            channel, _PS + "Classify", pb.ClassificationResponse)
+        self.Regress = _NativeProtoMethod(
+            channel, _PS + "Regress", pb.RegressionResponse)
+        self.MultiInference = _NativeProtoMethod(
+            channel, _PS + "MultiInference", pb.MultiInferenceResponse)
+        self.GetModelMetadata = _NativeProtoMethod(
+            channel, _PS + "GetModelMetadata", pb.GetModelMetadataResponse)
+
+
+class NativeModelServiceStub:
+    def __init__(self, channel):
+        from .wire import messages as pb
+        _MS = "/tensorflow.serving.ModelService/"
+        self.GetModelStatus = _NativeProtoMethod(
+            channel, _MS + "GetModelStatus", pb.GetModelStatusResponse)
+        self.HandleReloadConfigRequest = _NativeProtoMethod(
+            channel, _MS + "HandleReloadConfigRequest",
+            pb.ReloadConfigResponse)
+
+
 class TurboPredictClient:
     """Raw-bytes Predict client over the C++ codec.
 

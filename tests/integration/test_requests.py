@@ -272,3 +272,27 @@ def test_decode_predict_response_helper(client):
     np.testing.assert_array_equal(outs["x"], x)
     touts = decode_predict_response(resp, as_numpy=False)
     assert torch.equal(touts["x"], torch.arange(4, dtype=torch.float32))
+
+
+def test_client_backend_selection(server):
+    """TensorServingClient rides the C++ transport by default and falls
+    back to grpcio whenever grpcio-specific features are requested."""
+    host, port = "127.0.0.1", server.port
+    with TensorServingClient(host=host, port=port) as c:
+        assert c.backend == "native"
+        r = c.predict_request("default", {
+            "float_input": np.float32(np.random.rand(2, 3))})
+        assert "float_output" in r.outputs
+    with TensorServingClient(host=host, port=port,
+                             backend="grpcio") as c:
+        assert c.backend == "grpcio"
+        r = c.predict_request("default", {
+            "float_input": np.float32(np.random.rand(2, 3))})
+        assert "float_output" in r.outputs
+    with TensorServingClient(host=host, port=port,
+                             enable_retries=True) as c:
+        assert c.backend == "grpcio"  # retry policy is grpcio-specific
+    import pytest as _pytest
+    with _pytest.raises(ValueError):
+        TensorServingClient(host=host, port=port, backend="native",
+                            enable_retries=True)
