@@ -905,10 +905,35 @@ class GrpcServer {
   std::atomic<bool> stopping_{false};
   std::vector<std::thread> accept_threads_;
   std::vector<std::thread> workers_;
-  std::vector<std::thread> conn_threads_;
+  // connections are keyed so finished ones can be REAPED while the
+  // server runs: a churning client must not accumulate dead fds and
+  // threads until EMFILE (found by tools/soak_churn.py)
   std::mutex conns_mu_;
-  std::vector<std::shared_ptr<Conn>> conns_;
+  uint64_t conn_serial_ = 0;
+  std::map<uint64_t, std::thread> conn_threads_;
+  std::map<uint64_t, std::shared_ptr<Conn>> conns_;
+  std::deque<uint64_t> finished_conns_;
   std::vector<std::string> unix_paths_;  // unlink on stop
+
+  // joins connection threads whose loops have exited (called from the
+  // accept loops; never called for the caller's own thread)
+  void reap_finished_conns() {
+    std::vector<std::thread> done;
+    {
+      std::lock_guard<std::mutex> lk(conns_mu_);
+      while (!finished_conns_.empty()) {
+        uint64_t id = finished_conns_.front();
+        finished_conns_.pop_front();
+        auto it = conn_threads_.find(id);
+        if (it != conn_threads_.end()) {
+          done.push_back(std::move(it->second));
+          conn_threads_.erase(it);
+        }
+      }
+    }
+    for (auto& th : done)
+      if (th.joinable()) th.join();
+  }
 
   std::mutex handler_mu_;
   std::unordered_map<std::string, py::object> py_handlers_;
@@ -992,19 +1017,21 @@ class GrpcServer {
       ::setsockopt(cfd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
       h2::tune_socket(cfd);
       auto conn = std::make_shared<Conn>(cfd);
+      reap_finished_conns();
       {
         std::lock_guard<std::mutex> lk(conns_mu_);
         if (stopping_) {
           break;
         }
-        conns_.push_back(conn);
-        conn_threads_.emplace_back(
-            [this, conn] { connection_loop(conn); });
+        uint64_t id = ++conn_serial_;
+        conns_[id] = conn;
+        conn_threads_[id] = std::thread(
+            [this, conn, id] { connection_loop(conn, id); });
       }
     }
   }
 
-  void connection_loop(std::shared_ptr<Conn> conn) {
+  void connection_loop(std::shared_ptr<Conn> conn, uint64_t serial) {
     struct SrvStream {
       std::shared_ptr<CopyStream> stream_ref;  // must outlive dparse+body
       std::string path;
@@ -1133,6 +1160,11 @@ class GrpcServer {
       }
     } catch (const std::exception& e) {
       conn->mark_broken(e.what());
+    }
+    {
+      std::lock_guard<std::mutex> lk(conns_mu_);
+      conns_.erase(serial);  // fd closes when the last Task releases it
+      finished_conns_.push_back(serial);
     }
   }
 
@@ -1282,6 +1314,12 @@ class GrpcServer {
     bool have_parse = parsed_path && t.dparse.enabled &&
                       !t.dparse.failed && t.dparse.done;
     t.dparse.quiesce();  // drain copies even for failed/partial prospects
+    if (!Py_IsInitialized() || _Py_IsFinalizing()) {
+      // interpreter going down (abnormal teardown): acquiring the GIL
+      // now is fatal; shed the request instead
+      send_error_response(t, GRPC_UNAVAILABLE, "server shutting down");
+      return;
+    }
     int err_code = 0;
     std::string err_msg;
     py::object result;
@@ -1410,7 +1448,7 @@ class GrpcServer {
     listen_fds_.clear();
     {
       std::lock_guard<std::mutex> lk(conns_mu_);
-      for (auto& c : conns_) c->mark_broken("server stopping");
+      for (auto& kv : conns_) kv.second->mark_broken("server stopping");
     }
     q_cv_.notify_all();
     if (!was_running && !wait) return;
@@ -1418,11 +1456,16 @@ class GrpcServer {
       if (th.joinable()) th.join();
     accept_threads_.clear();
     {
+      std::map<uint64_t, std::thread> threads;
+      {
+        std::lock_guard<std::mutex> lk(conns_mu_);
+        threads.swap(conn_threads_);
+      }
+      for (auto& kv : threads)
+        if (kv.second.joinable()) kv.second.join();
       std::lock_guard<std::mutex> lk(conns_mu_);
-      for (auto& th : conn_threads_)
-        if (th.joinable()) th.join();
-      conn_threads_.clear();
       conns_.clear();
+      finished_conns_.clear();
     }
     for (auto& w : workers_)
       if (w.joinable()) w.join();

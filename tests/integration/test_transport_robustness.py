@@ -180,3 +180,28 @@ def test_huge_declared_message_rejected(server):
     time.sleep(0.2)
     s.close()
     _control_request_ok(addr)
+
+
+def test_connection_churn_releases_fds(server):
+    """Dead connections must be reaped while the server runs: 300
+    connect/close cycles may not grow this process's fd count
+    (regression for the EMFILE hang found by tools/soak_churn.py)."""
+    host, port, addr = server
+
+    def nfds():
+        return len(os.listdir("/proc/self/fd"))
+
+    # settle, then measure
+    for _ in range(10):
+        _raw(host, port).close()
+    time.sleep(0.3)
+    before = nfds()
+    for i in range(300):
+        s = _raw(host, port)
+        if i % 2 == 0:
+            s.sendall(PREFACE)
+        s.close()
+    time.sleep(1.0)
+    after = nfds()
+    assert after - before < 40, (before, after)
+    _control_request_ok(addr)
