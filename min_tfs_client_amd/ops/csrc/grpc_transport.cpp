@@ -256,16 +256,45 @@ inline bool tf_dtype_to_scalar(int dt, at::ScalarType* st) {
 
 struct MsgAssemblyFwd;  // below
 
-// Shared holder for a prospecting copy stream: the last owner (the
-// connection loop or a still-queued Task) quiesces and destroys it.
-struct CopyStream {
-  hipStream_t s = nullptr;
-  ~CopyStream() {
-    if (s != nullptr) {
-      (void)hipStreamSynchronize(s);
-      (void)hipStreamDestroy(s);
-    }
+// Small immortal pool of prospect copy streams per device. Connections
+// and channels BORROW a stream round-robin instead of creating their
+// own: per-connection hipStreamCreate/Destroy churn (thousands per
+// minute under client churn) stresses the driver — a box died under
+// tools/soak_churn.py before this pool existed. Streams are never
+// destroyed, so DeviceParse::quiesce can always sync its handle.
+class ProspectStreams {
+ public:
+  static ProspectStreams& instance() {
+    static ProspectStreams* p = new ProspectStreams();  // leaked: immortal
+    return *p;
   }
+
+  static constexpr int kPerDevice = 4;
+
+  hipStream_t get(int device) {
+    std::lock_guard<std::mutex> lk(mu_);
+    auto it = per_dev_.find(device);
+    if (it == per_dev_.end()) {
+      std::vector<hipStream_t> v;
+      if (hipSetDevice(device) == hipSuccess) {
+        for (int i = 0; i < kPerDevice; ++i) {
+          hipStream_t s = nullptr;
+          if (hipStreamCreateWithFlags(&s, hipStreamNonBlocking) !=
+              hipSuccess)
+            break;
+          v.push_back(s);
+        }
+      }
+      it = per_dev_.emplace(device, std::move(v)).first;
+    }
+    if (it->second.empty()) return nullptr;
+    return it->second[rr_++ % it->second.size()];
+  }
+
+ private:
+  std::mutex mu_;
+  std::unordered_map<int, std::vector<hipStream_t>> per_dev_;
+  size_t rr_ = 0;
 };
 
 struct DeviceParse {
@@ -942,7 +971,6 @@ class GrpcServer {
       echo_models_;
 
   struct Task {
-    std::shared_ptr<CopyStream> stream_ref;  // must outlive dparse+msg
     std::shared_ptr<Conn> conn;
     uint32_t stream = 0;
     std::string path;
@@ -1033,15 +1061,12 @@ class GrpcServer {
 
   void connection_loop(std::shared_ptr<Conn> conn, uint64_t serial) {
     struct SrvStream {
-      std::shared_ptr<CopyStream> stream_ref;  // must outlive dparse+body
       std::string path;
       HeaderBlock hb;
       bool headers_done = false;
       MsgAssembly body;
       DeviceParse dparse;  // dtor quiesces before body's buffer is freed
     };
-    // lazy per-connection copy stream for request prospecting
-    std::shared_ptr<CopyStream> conn_stream;
     std::unordered_map<uint32_t, SrvStream> streams;
     h2::HpackDecoder decoder;
     uint32_t continuation_stream = 0;
@@ -1086,7 +1111,7 @@ class GrpcServer {
             read_headers_fragment(conn->fd, fh, &st.hb.block);
             st.hb.end_stream = (fh.flags & h2::FL_END_STREAM) != 0;
             if (fh.flags & h2::FL_END_HEADERS) {
-              finish_headers(conn, fh.stream, st, decoder, &conn_stream);
+              finish_headers(conn, fh.stream, st, decoder);
               if (st.hb.end_stream) {
                 dispatch(conn, fh.stream, st);
                 streams.erase(fh.stream);
@@ -1106,7 +1131,7 @@ class GrpcServer {
                           reinterpret_cast<uint8_t*>(&st.hb.block[off]),
                           fh.length);
             if (fh.flags & h2::FL_END_HEADERS) {
-              finish_headers(conn, continuation_stream, st, decoder, &conn_stream);
+              finish_headers(conn, continuation_stream, st, decoder);
               if (st.hb.end_stream) {
                 dispatch(conn, continuation_stream, st);
                 streams.erase(continuation_stream);
@@ -1170,8 +1195,7 @@ class GrpcServer {
 
   template <typename SrvStreamT>
   void finish_headers(const std::shared_ptr<Conn>& conn, uint32_t stream,
-                      SrvStreamT& st, h2::HpackDecoder& decoder,
-                      std::shared_ptr<CopyStream>* conn_stream) {
+                      SrvStreamT& st, h2::HpackDecoder& decoder) {
     auto headers = decoder.decode(
         reinterpret_cast<const uint8_t*>(st.hb.block.data()),
         st.hb.block.size());
@@ -1188,18 +1212,11 @@ class GrpcServer {
       if (it != parsed_paths_.end()) device = it->second;
     }
     if (device >= 0) {
-      if (*conn_stream == nullptr) {
-        auto holder = std::make_shared<CopyStream>();
-        if (hipSetDevice(device) == hipSuccess &&
-            hipStreamCreateWithFlags(&holder->s, hipStreamNonBlocking) ==
-                hipSuccess)
-          *conn_stream = std::move(holder);
-      }
-      if (*conn_stream != nullptr) {
-        st.stream_ref = *conn_stream;
+      hipStream_t ps = ProspectStreams::instance().get(device);
+      if (ps != nullptr) {
         st.dparse.enabled = true;
         st.dparse.device = device;
-        st.dparse.stream = (*conn_stream)->s;
+        st.dparse.stream = ps;
         st.dparse.spec_field = 1;  // request layout
         st.dparse.map_field = 2;
         st.dparse.skip_model = [this](const std::string& name) {
@@ -1221,7 +1238,6 @@ class GrpcServer {
     t.path = std::move(st.path);
     t.msg = std::move(st.body.msg);
     t.dparse = std::move(st.dparse);
-    t.stream_ref = std::move(st.stream_ref);
     if (!st.body.have_len) {
       t.msg.alloc(0);  // empty request message (e.g. empty proto)
       t.msg.len = 0;
@@ -1567,19 +1583,10 @@ class GrpcChannel {
   // streaming variant: the request payload is a skeleton plus regions
   // (see write_message_with_regions) — device regions overlap DMA with
   // the send, host regions go out zero-copy straight from tensor memory
-  // lazily created copy stream for receive-side device parse; nullptr
-  // when a different device was already bound (caller falls back)
+  // borrowed immortal prospect stream (see ProspectStreams); nullptr
+  // when the device is unavailable (caller falls back to plain receive)
   hipStream_t device_stream(int dev) {
-    std::lock_guard<std::mutex> lk(dstream_mu_);
-    if (dstream_ != nullptr) return dstream_dev_ == dev ? dstream_ : nullptr;
-    if (hipSetDevice(dev) != hipSuccess) return nullptr;
-    if (hipStreamCreateWithFlags(&dstream_, hipStreamNonBlocking) !=
-        hipSuccess) {
-      dstream_ = nullptr;
-      return nullptr;
-    }
-    dstream_dev_ = dev;
-    return dstream_;
+    return ProspectStreams::instance().get(dev);
   }
 
   uint32_t start_call_streaming(const std::string& path, const uint8_t* buf,
@@ -1740,10 +1747,6 @@ class GrpcChannel {
 
   std::shared_ptr<Conn> conn_;
   std::thread reader_;
-  // receive-side device-parse copy stream (lazy; one per channel)
-  std::mutex dstream_mu_;
-  hipStream_t dstream_ = nullptr;
-  int dstream_dev_ = -1;
   std::string authority_;
   std::atomic<bool> closed_{false};
   std::mutex mu_;
