@@ -20,6 +20,10 @@
 #include <pybind11/stl.h>
 #include <torch/extension.h>
 
+#include <c10/hip/HIPCachingAllocator.h>
+#include <c10/hip/HIPGuard.h>
+#include <c10/hip/HIPStream.h>
+
 #include <algorithm>
 #include <atomic>
 #include <chrono>
@@ -252,6 +256,21 @@ inline bool tf_dtype_to_scalar(int dt, at::ScalarType* st) {
     case 23: *st = at::kUInt64; return true;
     default: return false;
   }
+}
+
+
+// Record the calling thread's current stream as a user of every
+// prospected tensor: their blocks are associated with the pool copy
+// stream, so without this the allocator could hand a freed block back
+// to the copy stream while the consumer's reads are still in flight.
+inline void record_consumer_stream(
+    std::vector<std::pair<std::string, at::Tensor>>& outs, int device) {
+  if (outs.empty()) return;
+  auto cur = c10::hip::getCurrentHIPStream(
+      static_cast<c10::DeviceIndex>(device));
+  for (auto& kv : outs)
+    c10::hip::HIPCachingAllocator::recordStream(
+        kv.second.storage().data_ptr(), cur);
 }
 
 struct MsgAssemblyFwd;  // below
@@ -585,6 +604,8 @@ inline void device_parse_advance(DeviceParse& dp, MsgAssembly& body) {
               for (int64_t d : dp.cur_shape) numel *= d;
               at::ScalarType st;
               if (numel == 0 && tf_dtype_to_scalar(dp.cur_dtype, &st)) {
+                c10::hip::HIPStreamGuard g(c10::hip::getStreamFromExternal(
+                    dp.stream, static_cast<c10::DeviceIndex>(dp.device)));
                 dp.cur_tensor = at::empty(
                     dp.cur_shape, at::TensorOptions().dtype(st).device(
                                       at::kCUDA, dp.device));
@@ -641,9 +662,18 @@ inline void device_parse_advance(DeviceParse& dp, MsgAssembly& body) {
             for (int64_t d : dp.cur_shape) numel *= d;
             size_t esize = at::elementSize(st);
             if (uint64_t(numel) * esize != len) { dp.fail(); return; }
-            dp.cur_tensor = at::empty(
-                dp.cur_shape,
-                at::TensorOptions().dtype(st).device(at::kCUDA, dp.device));
+            {
+              // allocate ON the copy stream: the caching allocator then
+              // orders block reuse against S, so our H2D never lands in
+              // a block another stream is still reading (the cross-
+              // stream reuse race tools/soak_churn.py exposed)
+              c10::hip::HIPStreamGuard g(c10::hip::getStreamFromExternal(
+                  dp.stream, static_cast<c10::DeviceIndex>(dp.device)));
+              dp.cur_tensor = at::empty(
+                  dp.cur_shape,
+                  at::TensorOptions().dtype(st).device(at::kCUDA,
+                                                       dp.device));
+            }
             dp.content_off = off;
             dp.content_len = size_t(len);
             dp.content_copied = 0;
@@ -1348,6 +1378,7 @@ class GrpcServer {
           py::object spec = py::none();
           py::object outs = py::none();
           if (have_parse) {
+            record_consumer_stream(t.dparse.outs, t.dparse.device);
             py::dict sd;
             sd["name"] = t.dparse.spec_name;
             sd["version"] = t.dparse.spec_version;
@@ -1674,6 +1705,8 @@ class GrpcChannel {
     hipStream_t ds = p->dparse.stream;
     lk.unlock();
     if (need_sync && hipStreamSynchronize(ds) != hipSuccess) parsed = false;
+    if (parsed) record_consumer_stream(r.outs, r.outs.empty() ? 0 :
+        r.outs.front().second.get_device());
     {
       std::lock_guard<std::mutex> glk(mu_);
       pending_.erase(id);

@@ -45,7 +45,7 @@ transport_ext = CppExtension(
     sources=[os.path.join(_CSRC, "grpc_transport.cpp")],
     include_dirs=[os.path.join(_ROCM, "include")],
     library_dirs=[os.path.join(_ROCM, "lib")],
-    libraries=["amdhip64"],
+    libraries=["amdhip64", "c10_hip", "torch_hip"],
     extra_compile_args={"cxx": ["-O3", "-std=c++17"]},
 )
 
