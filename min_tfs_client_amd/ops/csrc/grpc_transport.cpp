@@ -360,9 +360,8 @@ struct DeviceParse {
   // source buffer and the destination tensors are freed — a pinned
   // receive buffer hipHostFree'd with a pending SDMA read (or a tensor
   // freed with a pending write) is a crash/corruption. The dtor body
-  // runs before member (outs) destruction, and holders declare the
-  // stream's keepalive BEFORE the DeviceParse member so the stream
-  // outlives this sync.
+  // runs before member (outs) destruction; the stream itself is an
+  // immortal ProspectStreams handle, so syncing it here is always valid.
   void quiesce() {
     if (copies_issued && stream != nullptr) {
       (void)hipStreamSynchronize(stream);
@@ -1340,17 +1339,19 @@ class GrpcServer {
         }
       }
     }
-    // 2) python handler
-    py::object fn;
+    // 2) python handler. py::object copies/destruction are refcount ops
+    // and MUST happen under the GIL — only an existence check (count()
+    // touches no refcounts) outside it; the actual handler reference
+    // lives inside the GIL block below.
+    bool have_handler;
     {
       std::lock_guard<std::mutex> lk(handler_mu_);
-      auto it = py_handlers_.find(t.path);
-      if (it == py_handlers_.end()) {
-        send_error_response(t, GRPC_UNIMPLEMENTED,
-                            "unknown service method " + t.path);
-        return;
-      }
-      fn = it->second;
+      have_handler = py_handlers_.count(t.path) != 0;
+    }
+    if (!have_handler) {
+      send_error_response(t, GRPC_UNIMPLEMENTED,
+                          "unknown service method " + t.path);
+      return;
     }
     bool parsed_path;
     {
@@ -1371,6 +1372,16 @@ class GrpcServer {
     py::object result;
     {
       py::gil_scoped_acquire gil;
+      py::object fn;
+      {
+        std::lock_guard<std::mutex> lk(handler_mu_);
+        auto it = py_handlers_.find(t.path);
+        if (it != py_handlers_.end()) fn = it->second;
+      }
+      if (!fn) {
+        err_code = GRPC_UNIMPLEMENTED;
+        err_msg = "service method unregistered: " + t.path;
+      } else
       try {
         py::memoryview view = py::memoryview::from_memory(
             t.msg.p, py::ssize_t(t.msg.len));
