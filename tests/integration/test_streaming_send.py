@@ -239,3 +239,35 @@ def test_register_handler_parsed_no_gpu_fallback():
         assert calls and calls[0] == (None, None)
     finally:
         srv.stop()
+
+
+def test_call_streaming_parsed_deadline(tmp_path):
+    """Deadline on a parse-ahead call: DEADLINE_EXCEEDED surfaces, the
+    channel and server both stay usable."""
+    import time as _time
+
+    def slow(view):
+        _time.sleep(1.0)
+        return bytes(view)
+
+    srv = T.GrpcServer("127.0.0.1:0", 2)
+    srv.register_handler("/t.S/Slow", slow)
+    srv.register_handler("/t.S/Echo", lambda v: bytes(v))
+    addr = srv.start()
+    try:
+        inputs = {"x": torch.randn(512, 512)}
+        blob, regions, keep = _streaming_parts(inputs)
+        ch = T.GrpcChannel(addr)
+        try:
+            with pytest.raises(T.NativeRpcError) as ei:
+                ch.call_streaming_parsed("/t.S/Slow", blob, list(regions),
+                                         0, 0.05)
+            assert ei.value.code_int == 4  # DEADLINE_EXCEEDED
+            # channel still healthy afterwards
+            outs, raw = ch.call_streaming_parsed(
+                "/t.S/Echo", blob, list(regions), 0, 30.0)
+            assert bytes(raw) == _buffered(inputs)
+        finally:
+            ch.close()
+    finally:
+        srv.stop()
