@@ -332,3 +332,69 @@ def test_echo_model_skips_prospecting(tmp_path):
             # process here): allow client-side parse-ahead allocs only
             after = torch.cuda.memory_allocated()
             assert after - before < 200 << 20
+
+
+def test_parse_ahead_fuzz_random_responses():
+    """50 randomized canonical responses (mixed dtypes, scalars, empty
+    tensors, long names, multi-chunk payloads): parse-ahead must either
+    match the ordinary parse exactly or fall back to None."""
+    import random
+
+    rng = random.Random(20260914)
+    dtypes = [torch.float32, torch.float64, torch.int32, torch.int64,
+              torch.int16, torch.uint8, torch.bfloat16, torch.float16,
+              torch.bool]
+
+    srv = T.GrpcServer("127.0.0.1:0", 2)
+    payload_holder = {}
+    srv.register_handler("/t.S/Fuzz", lambda v: payload_holder["p"])
+    addr = srv.start()
+    try:
+        ch = T.GrpcChannel(addr)
+        req_blob, req_regions, keep = _streaming(
+            {"q": torch.zeros(4, device=DEV)})
+        try:
+            for it in range(50):
+                n_tensors = rng.randint(1, 5)
+                outputs = {}
+                for i in range(n_tensors):
+                    dt = rng.choice(dtypes)
+                    kind = rng.random()
+                    if kind < 0.15:
+                        shape = ()  # scalar
+                    elif kind < 0.3:
+                        shape = (0, rng.randint(1, 8))  # empty
+                    elif kind < 0.5:
+                        shape = (rng.randint(1, 2048),)
+                    else:
+                        shape = (rng.randint(1, 48), 3,
+                                 rng.randint(8, 224), rng.randint(8, 224))
+                    name = ("t" * rng.randint(1, 40)) + str(i)
+                    if dt == torch.bool:
+                        t = torch.randint(0, 2, shape, device=DEV
+                                          ).to(torch.bool)
+                    elif dt.is_floating_point:
+                        t = torch.randn(shape, device=DEV).to(dt)
+                    else:
+                        t = torch.randint(0, 100, shape, dtype=dt,
+                                          device=DEV)
+                    outputs[name] = t
+                payload_holder["p"] = _buffered(outputs, is_request=False)
+                outs, raw = ch.call_streaming_parsed(
+                    "/t.S/Fuzz", req_blob, list(req_regions), 0, 60.0)
+                _s, ref, _f = native.parse_predict_response(
+                    bytes(raw), "cuda:0", 1)
+                if outs is None:
+                    # acceptable only for sub-1MB messages (machinery
+                    # skips them); big canonical ones must prospect
+                    assert len(bytes(raw)) < (1 << 20), it
+                    outs = ref
+                assert set(outs) == set(ref), it
+                for k in ref:
+                    assert outs[k].dtype == ref[k].dtype, (it, k)
+                    assert outs[k].shape == ref[k].shape, (it, k)
+                    assert torch.equal(outs[k], ref[k]), (it, k)
+        finally:
+            ch.close()
+    finally:
+        srv.stop()
